@@ -1,0 +1,138 @@
+"""Seeded synthetic SSTable run generator for the BASELINE.json configs.
+
+Shapes (BASELINE.json `configs`):
+  cfg1: 2 runs x 10k entries, 16 B keys / 64 B values
+  cfg2: 4 runs x 1M  entries, 16 B keys / 256 B values
+  cfg3: 8 runs x ~987k entries (1 GiB data each), 32 B keys / 1 KiB values,
+        50% key overlap (half the keys appear in exactly 2 runs),
+        5% tombstones
+  cfg4: 64 jobs of (4 runs x 256 MiB), cfg2 shapes — built per-job via
+        make_job(..., job_seed)
+  cfg5: variable-length keys (later round)
+
+Determinism: default seed 0xDBEE1; timestamps strictly increase with run
+index (ts = run_index * 2^40 + position) so merge winners are deterministic
+(BASELINE.md). Keys are uniformly random bytes, sorted ascending per run,
+unique within a run (dbeel flush invariant, lsm_tree.rs:925-946).
+"""
+from __future__ import annotations
+
+import numpy as np
+
+from .format import build_run_fixed_key
+
+DEFAULT_SEED = 0xDBEE1
+TS_RUN_STRIDE = 1 << 40
+
+
+def _sort_fixed_keys(keys: np.ndarray) -> np.ndarray:
+    """Lexicographic ascending sort of (N, K) u8 keys, K multiple of 8."""
+    N, K = keys.shape
+    cols = keys.view(">u8").reshape(N, K // 8)  # big-endian u64 words
+    order = np.lexsort(tuple(cols[:, j] for j in range(K // 8 - 1, -1, -1)))
+    return keys[order]
+
+
+def _unique_keys(rng: np.random.Generator, n: int, K: int) -> np.ndarray:
+    """n distinct random K-byte keys (u8 matrix), not sorted."""
+    keys = rng.integers(0, 256, size=(n, K), dtype=np.uint8)
+    # collisions at 128+ random bits are ~impossible; dedupe defensively
+    v = np.unique(keys.view([("", "u1")] * K))
+    while v.shape[0] < n:
+        extra = rng.integers(0, 256, size=(n, K), dtype=np.uint8)
+        v = np.unique(
+            np.concatenate([v, extra.view([("", "u1")] * K).reshape(-1)])
+        )
+    arr = v.view(np.uint8).reshape(-1, K)[:n]
+    return rng.permutation(arr, axis=0)
+
+
+def make_runs(
+    n_runs: int,
+    entries_per_run: int,
+    key_size: int,
+    value_size: int,
+    overlap_frac: float = 0.0,
+    tombstone_frac: float = 0.0,
+    seed: int = DEFAULT_SEED,
+) -> list[tuple[np.ndarray, np.ndarray]]:
+    """Build n_runs (data, index) byte-array pairs.
+
+    overlap_frac o: a fraction o of each run's entries use keys shared with
+    exactly one other run (paired runs 2i/2i+1), so those keys appear in
+    exactly 2 runs; the rest are unique to the run.
+    tombstone_frac t: fraction of each run's entries with empty values.
+    """
+    rng = np.random.default_rng(seed)
+    n_shared = int(entries_per_run * overlap_frac)
+    n_own = entries_per_run - n_shared
+
+    # Draw one global pool of distinct keys, then partition: per-pair shared
+    # sets and per-run unique sets.
+    n_pairs = n_runs // 2
+    total = n_pairs * n_shared + n_runs * n_own + (n_runs % 2) * n_shared
+    pool = _unique_keys(rng, total, key_size)
+    pos = 0
+    shared_sets = []
+    for p in range(n_pairs):
+        shared_sets.append(pool[pos : pos + n_shared])
+        pos += n_shared
+    odd_shared = None
+    if n_runs % 2 and n_shared:
+        odd_shared = pool[pos : pos + n_shared]  # unpaired run: keys used once
+        pos += n_shared
+
+    out = []
+    for r in range(n_runs):
+        own = pool[pos : pos + n_own]
+        pos += n_own
+        if n_shared:
+            if r // 2 < n_pairs:
+                sh = shared_sets[r // 2]
+            else:
+                sh = odd_shared
+            keys = np.concatenate([own, sh], axis=0)
+        else:
+            keys = own
+        keys = _sort_fixed_keys(keys)
+        n = keys.shape[0]
+
+        vsizes = np.full(n, value_size, dtype=np.uint64)
+        if tombstone_frac > 0:
+            tomb = rng.random(n) < tombstone_frac
+            vsizes[tomb] = 0
+        n_val_bytes = int(vsizes.sum())
+        vfill = rng.integers(0, 256, size=n_val_bytes, dtype=np.uint8)
+        ts = np.uint64(r * TS_RUN_STRIDE) + np.arange(n, dtype=np.uint64)
+        data, index = build_run_fixed_key(keys, vsizes, vfill, ts)
+        out.append((data, index))
+    return out
+
+
+CONFIGS = {
+    "cfg1": dict(n_runs=2, entries_per_run=10_000, key_size=16, value_size=64),
+    "cfg2": dict(n_runs=4, entries_per_run=1_000_000, key_size=16, value_size=256),
+    "cfg3": dict(
+        n_runs=8,
+        entries_per_run=987_000,
+        key_size=32,
+        value_size=1024,
+        overlap_frac=0.5,
+        tombstone_frac=0.05,
+    ),
+    # cfg4 = 64 jobs of this shape (per-job seed via make_job)
+    "cfg4_job": dict(
+        n_runs=4, entries_per_run=883_000, key_size=16, value_size=256
+    ),
+}
+
+
+def make_config(name: str, seed: int = DEFAULT_SEED, scale: float = 1.0):
+    cfg = dict(CONFIGS[name])
+    if scale != 1.0:
+        cfg["entries_per_run"] = max(16, int(cfg["entries_per_run"] * scale))
+    return make_runs(seed=seed, **cfg)
+
+
+def input_bytes(runs) -> int:
+    return sum(len(d) + len(i) for d, i in runs)

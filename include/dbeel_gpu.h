@@ -1,0 +1,133 @@
+/* dbeel_gpu.h — C ABI of the MI355X-native SSTable compaction engine.
+ *
+ * This is the drop-in seam for dbeel's `LSMTree::compact` hot path
+ * (reference: src/storage_engine/lsm_tree.rs:950-1156). Everything from
+ * lsm_tree.rs:974 down (read -> k-way merge/dedup -> write) is replaced by
+ * `dbeel_gpu_compact`; the caller keeps the reference's surrounding side
+ * effects (bloom file, .compact_action journal, renames, sstable-list swap —
+ * lsm_tree.rs:1070-1153).
+ *
+ * Wire format (all integers little-endian, bincode fixint — reference
+ * src/utils/bincode.rs:9-16):
+ *   entry      = key_len:u64 | key bytes | data_len:u64 | data bytes |
+ *                timestamp:i128                 (src/storage_engine/mod.rs:68-73,
+ *                                               src/utils/timestamp_nanos.rs:6-11)
+ *   index rec  = offset:u64 | key_size:u32 | full_size:u32   = 16 bytes
+ *                (src/storage_engine/mod.rs:45-50, INDEX_ENTRY_SIZE mod.rs:33,
+ *                 asserted lsm_tree.rs:408-413)
+ *   key_size   = 8 + key_len;  full_size = 32 + key_len + data_len
+ *
+ * Merge semantics (lsm_tree.rs:1038-1066, CompactionItem ordering
+ * lsm_tree.rs:52-71 + Entry::cmp mod.rs:75-81):
+ *   total order: key bytes ascending (lexicographic), then timestamp
+ *   ascending (signed i128), then run index ascending. Among entries with
+ *   equal key, exactly one survives: the max by (timestamp, run index)
+ *   ("last pop of an equal-key group wins"). A surviving entry whose data is
+ *   empty (TOMBSTONE, mod.rs:14) is dropped unless keep_tombstones.
+ *   Survivors are emitted verbatim (input bytes unchanged) in total order,
+ *   densely packed; index offsets recomputed from 0
+ *   (entry_writer.rs:71-98).
+ *
+ * Input invariants (guaranteed by dbeel's flush/compaction writers,
+ * lsm_tree.rs:925-946): each run is sorted ascending by key with unique keys
+ * within the run. Corrupt inputs (index size not a multiple of 16, entry
+ * ranges out of bounds, full_size < 32 + key_len) return
+ * DBEEL_ERR_CORRUPT_RECORD; the reference silently treats a failed decode as
+ * end-of-stream (lsm_tree.rs:1014,1063 "if let Ok"), this engine errors
+ * loudly instead (documented divergence).
+ */
+#ifndef DBEEL_GPU_H
+#define DBEEL_GPU_H
+
+#include <stddef.h>
+#include <stdint.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+/* One SSTable run, resident in host DRAM, caller-owned, read-only.
+ * runs[] must be ordered by ascending sstable index — the (timestamp, run
+ * index) tie-break depends on it (lsm_tree.rs:58-65). */
+typedef struct {
+    const uint8_t* data;      /* run data file bytes (concatenated entries) */
+    size_t         data_len;
+    const uint8_t* index;     /* run index file bytes (16-byte records)     */
+    size_t         index_len;
+} dbeel_run_view;
+
+/* Engine-allocated outputs; free with dbeel_gpu_result_free. */
+typedef struct {
+    uint8_t* data;            /* output .data file bytes  */
+    size_t   data_len;
+    uint8_t* index;           /* output .index file bytes */
+    size_t   index_len;
+    uint64_t entries_written;
+} dbeel_compact_result;
+
+/* Optional per-call timing/throughput evidence (milliseconds, HIP events on
+ * the job's stream). kernel_ms covers the whole device pipeline
+ * (rank -> scan -> emit -> copy), excluding host<->device copies. */
+typedef struct {
+    double h2d_ms;
+    double rank_ms;    /* k_rank: global merge-rank + winner flags  */
+    double scan_ms;    /* size/position prefix sums                 */
+    double emit_ms;    /* output index build + survivor source map  */
+    double copy_ms;    /* verbatim entry copy-out                   */
+    double kernel_ms;  /* rank+scan+emit+copy                       */
+    double d2h_ms;
+} dbeel_compact_timings;
+
+/* Error codes. Mirrors the reference error taxonomy at this seam
+ * (src/error.rs; ItemTooLarge error.rs:60-61). */
+enum {
+    DBEEL_OK               = 0,
+    DBEEL_ERR_INVALID_ARG  = 1,
+    DBEEL_ERR_CORRUPT      = 2,  /* bincode-decode-failure analogue */
+    DBEEL_ERR_ITEM_TOO_LARGE = 3,
+    DBEEL_ERR_HIP          = 4,  /* device/runtime failure (I/O analogue) */
+    DBEEL_ERR_NO_GPU       = 5,
+    DBEEL_ERR_OOM          = 6,
+};
+
+/* Compact n_runs runs into one run. keep_tombstones: 0 = drop entries with
+ * empty value data (lsm_tree.rs:1045-1046). device: HIP device ordinal
+ * (>= 0). There is NO CPU fallback in this library: the CPU restatement
+ * lives in liboracle.so (test infrastructure only) and device < 0 returns
+ * DBEEL_ERR_INVALID_ARG. Returns 0 on success. Reentrant; concurrent calls
+ * on distinct devices run independent jobs. */
+int dbeel_gpu_compact(const dbeel_run_view* runs, size_t n_runs,
+                      int keep_tombstones, int device,
+                      dbeel_compact_result* out);
+
+/* Same, and also reports timings (t may be NULL). */
+int dbeel_gpu_compact_timed(const dbeel_run_view* runs, size_t n_runs,
+                            int keep_tombstones, int device,
+                            dbeel_compact_result* out,
+                            dbeel_compact_timings* t);
+
+void dbeel_gpu_result_free(dbeel_compact_result* r);
+
+/* Thread-local message for the last error in this thread. */
+const char* dbeel_gpu_last_error(void);
+
+/* ---- Resident-job API (benchmarking / repeated compactions) ----
+ * Uploads the runs to `device` once; each run executes the device pipeline
+ * with inputs already resident in HBM (what BASELINE's MB/s is quoted on)
+ * and leaves outputs on the device. */
+typedef struct dbeel_gpu_job dbeel_gpu_job;
+
+int dbeel_gpu_job_create(const dbeel_run_view* runs, size_t n_runs,
+                         int device, dbeel_gpu_job** out_job);
+/* Runs the pipeline; fills t (may be NULL) and the output sizes. */
+int dbeel_gpu_job_run(dbeel_gpu_job* job, int keep_tombstones,
+                      uint64_t* out_data_len, uint64_t* out_entries,
+                      dbeel_compact_timings* t);
+/* Copies the last run's outputs to host (engine-allocated). */
+int dbeel_gpu_job_fetch(dbeel_gpu_job* job, dbeel_compact_result* out);
+void dbeel_gpu_job_destroy(dbeel_gpu_job* job);
+
+#ifdef __cplusplus
+}
+#endif
+#endif /* DBEEL_GPU_H */
