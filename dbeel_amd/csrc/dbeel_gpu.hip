@@ -1,0 +1,762 @@
+/* dbeel_gpu.hip — MI355X-native (gfx950/CDNA4) SSTable compaction engine.
+ *
+ * Implements include/dbeel_gpu.h: the drop-in replacement for dbeel's
+ * LSMTree::compact hot loop (reference lsm_tree.rs:950-1156; semantics
+ * restated in the header). This is a from-scratch GPU design, not a port of
+ * the reference's heap loop:
+ *
+ *   k_validate : each run strictly sorted by key, entries well-formed
+ *                (dbeel flush invariant, lsm_tree.rs:925-946; corrupt input
+ *                errors loudly where the reference silently truncates)
+ *   k_rank     : every entry's global merge rank by per-run binary search
+ *                (order = key bytes asc, timestamp i128 asc, run index asc —
+ *                lsm_tree.rs:52-71 + mod.rs:75-81); the winner test
+ *                ("no other run holds an equal key later in the order",
+ *                lsm_tree.rs:1041-1044) falls out of the same searches free.
+ *   scans      : rocPRIM exclusive scans of survivor sizes/flags ->
+ *                output byte offsets + positions.
+ *   k_emit     : output .index records (offset/key_size/full_size,
+ *                entry_writer.rs:79-87) + compacted source map.
+ *   k_copy     : verbatim survivor byte copy, balanced by DESTINATION
+ *                granule (one 16-B granule per lane, 4-KiB window per
+ *                256-thread block) so throughput is independent of entry
+ *                size; aligned 16-B stores, unaligned 16-B loads.
+ *
+ * All integer/byte work, HBM-bandwidth bound; MFMA unused by design
+ * (BASELINE.json north_star).
+ *
+ * Build: hipcc --offload-arch=gfx950 -O3 -shared -fPIC dbeel_gpu.hip
+ *        -o libdbeel_gpu.so
+ */
+#include <cstdarg>
+#include <cstdint>
+#include <cstdio>
+#include <cstring>
+#include <cstdlib>
+#include <new>
+
+#include <hip/hip_runtime.h>
+#include <rocprim/rocprim.hpp>
+
+#include "../../include/dbeel_gpu.h"
+
+#define MAX_RUNS 64
+
+static thread_local char g_err[512];
+
+static void set_err(const char* fmt, ...) {
+    va_list ap;
+    va_start(ap, fmt);
+    vsnprintf(g_err, sizeof g_err, fmt, ap);
+    va_end(ap);
+}
+
+extern "C" const char* dbeel_gpu_last_error(void) { return g_err; }
+
+#define HIP_CHECK(call)                                                     \
+    do {                                                                    \
+        hipError_t _e = (call);                                             \
+        if (_e != hipSuccess) {                                             \
+            set_err("%s failed: %s (%s:%d)", #call, hipGetErrorString(_e),  \
+                    __FILE__, __LINE__);                                    \
+            return (_e == hipErrorOutOfMemory) ? DBEEL_ERR_OOM              \
+                                               : DBEEL_ERR_HIP;             \
+        }                                                                   \
+    } while (0)
+
+/* ------------------------------------------------------------------ */
+/* Device-side format view                                            */
+/* ------------------------------------------------------------------ */
+
+struct RunsDesc {
+    const uint8_t* data[MAX_RUNS];
+    const uint8_t* index[MAX_RUNS];
+    uint64_t data_len[MAX_RUNS];
+    uint64_t count[MAX_RUNS];       /* entries per run                  */
+    uint64_t entry_base[MAX_RUNS];  /* exclusive prefix sum of count    */
+    int n_runs;
+    uint64_t total;                 /* sum of count                     */
+};
+
+/* 16 B/entry scratch record, scattered by global rank in k_rank. */
+struct EntryMeta {
+    uint64_t src;        /* run (high 16 bits) | data offset (low 48)  */
+    uint32_t key_size;   /* 8 + key_len (index record field)           */
+    uint32_t full_size;
+};
+
+struct EView {
+    const uint8_t* raw;
+    const uint8_t* key;
+    uint64_t klen;
+    uint64_t off;
+    uint32_t key_size, full_size;
+};
+
+__device__ __forceinline__ uint64_t ld_u64(const uint8_t* p) {
+    uint64_t v;
+    __builtin_memcpy(&v, p, 8);
+    return v;
+}
+__device__ __forceinline__ uint32_t ld_u32(const uint8_t* p) {
+    uint32_t v;
+    __builtin_memcpy(&v, p, 4);
+    return v;
+}
+
+/* Loads index record + key view. Returns false on corrupt record. */
+__device__ __forceinline__ bool load_entry(const RunsDesc& R, int r,
+                                           uint64_t i, EView& e) {
+    const uint8_t* rec = R.index[r] + i * 16;
+    e.off = ld_u64(rec);
+    e.key_size = ld_u32(rec + 8);
+    e.full_size = ld_u32(rec + 12);
+    if (e.key_size < 8 || e.full_size < 32 ||
+        (uint64_t)e.key_size + 24 > e.full_size ||
+        e.off + e.full_size > R.data_len[r])
+        return false;
+    e.raw = R.data[r] + e.off;
+    e.klen = e.key_size - 8;
+    e.key = e.raw + 8;
+    return true;
+}
+
+/* Lexicographic compare of key byte strings (Vec<u8> cmp, mod.rs:75-81). */
+__device__ __forceinline__ int cmp_keys(const uint8_t* a, uint64_t la,
+                                        const uint8_t* b, uint64_t lb) {
+    uint64_t n = la < lb ? la : lb;
+    uint64_t i = 0;
+    for (; i + 8 <= n; i += 8) {
+        uint64_t va = ld_u64(a + i), vb = ld_u64(b + i);
+        if (va != vb) {
+            va = __builtin_bswap64(va);
+            vb = __builtin_bswap64(vb);
+            return va < vb ? -1 : 1;
+        }
+    }
+    if (i < n) {
+        uint64_t rem = n - i;
+        uint64_t va = 0, vb = 0;
+        for (uint64_t j = 0; j < rem; j++) {  /* <= 7 bytes */
+            va |= (uint64_t)a[i + j] << (8 * j);
+            vb |= (uint64_t)b[i + j] << (8 * j);
+        }
+        if (va != vb) {
+            va = __builtin_bswap64(va);
+            vb = __builtin_bswap64(vb);
+            return va < vb ? -1 : 1;
+        }
+    }
+    return la < lb ? -1 : (la > lb ? 1 : 0);
+}
+
+/* Timestamp: trailing i128 LE (utils/timestamp_nanos.rs:6-11). */
+__device__ __forceinline__ void load_ts(const EView& e, uint64_t& lo,
+                                        int64_t& hi) {
+    lo = ld_u64(e.raw + e.full_size - 16);
+    int64_t h;
+    __builtin_memcpy(&h, e.raw + e.full_size - 8, 8);
+    hi = h;
+}
+
+/* Full total order: (key, timestamp, run index) — lsm_tree.rs:52-71. */
+__device__ __forceinline__ int cmp_full(const EView& a, int ra,
+                                        const EView& b, int rb) {
+    int c = cmp_keys(a.key, a.klen, b.key, b.klen);
+    if (c) return c;
+    uint64_t alo, blo;
+    int64_t ahi, bhi;
+    load_ts(a, alo, ahi);
+    load_ts(b, blo, bhi);
+    if (ahi != bhi) return ahi < bhi ? -1 : 1;
+    if (alo != blo) return alo < blo ? -1 : 1;
+    return ra < rb ? -1 : (ra > rb ? 1 : 0);
+}
+
+/* ------------------------------------------------------------------ */
+/* Kernels                                                            */
+/* ------------------------------------------------------------------ */
+
+/* err codes stored in the device flag */
+#define DERR_CORRUPT 1u
+#define DERR_UNSORTED 2u
+
+__global__ void k_validate(RunsDesc R, uint32_t* err) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t g = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         g < R.total; g += stride) {
+        /* locate run (<= 64 runs: linear scan on cached desc) */
+        int r = 0;
+        while (r + 1 < R.n_runs && g >= R.entry_base[r + 1]) r++;
+        uint64_t i = g - R.entry_base[r];
+        EView e;
+        if (!load_entry(R, r, i, e)) {
+            atomicOr(err, DERR_CORRUPT);
+            continue;
+        }
+        /* bincode field cross-check (read_next_entry, lsm_tree.rs:1158-70) */
+        if (ld_u64(e.raw) != e.klen ||
+            ld_u64(e.raw + 8 + e.klen) != (uint64_t)e.full_size - 32 - e.klen) {
+            atomicOr(err, DERR_CORRUPT);
+            continue;
+        }
+        /* run must be strictly sorted by key (flush invariant) */
+        if (i + 1 < R.count[r]) {
+            EView nx;
+            if (!load_entry(R, r, i + 1, nx)) {
+                atomicOr(err, DERR_CORRUPT);
+                continue;
+            }
+            if (cmp_keys(e.key, e.klen, nx.key, nx.klen) >= 0)
+                atomicOr(err, DERR_UNSORTED);
+        }
+    }
+}
+
+/* Number of entries in run r2 strictly below e (full order). Runs are
+ * sorted by key with unique keys, hence sorted in full order too. */
+__device__ __forceinline__ uint64_t lower_rank(const RunsDesc& R, int r2,
+                                               const EView& e, int re,
+                                               bool& equal_key_at) {
+    uint64_t lo = 0, hi = R.count[r2];
+    while (lo < hi) {
+        uint64_t mid = (lo + hi) >> 1;
+        EView m;
+        /* corrupt records are flagged by k_validate; avoid the OOB read
+         * here (results are discarded once the host sees the flag) */
+        if (!load_entry(R, r2, mid, m)) {
+            hi = mid;
+            continue;
+        }
+        if (cmp_full(m, r2, e, re) < 0)
+            lo = mid + 1;
+        else
+            hi = mid;
+    }
+    equal_key_at = false;
+    if (lo < R.count[r2]) {
+        EView m;
+        if (load_entry(R, r2, lo, m))
+            equal_key_at = (m.klen == e.klen) &&
+                           (cmp_keys(m.key, m.klen, e.key, e.klen) == 0);
+    }
+    return lo;
+}
+
+__global__ void k_rank(RunsDesc R, EntryMeta* meta, uint64_t* sizes,
+                       uint32_t* flags, int keep_tombstones) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t g = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         g < R.total; g += stride) {
+        int r = 0;
+        while (r + 1 < R.n_runs && g >= R.entry_base[r + 1]) r++;
+        uint64_t i = g - R.entry_base[r];
+        EView e;
+        if (!load_entry(R, r, i, e)) {
+            /* k_validate has flagged this input; keep memory safe and park
+             * the entry at its local slot (results will be discarded) */
+            meta[g] = EntryMeta{0, 8, 32};
+            sizes[g] = 0;
+            flags[g] = 0;
+            continue;
+        }
+
+        uint64_t rank = i;
+        bool winner = true;
+        for (int r2 = 0; r2 < R.n_runs; r2++) {
+            if (r2 == r) continue;
+            bool eq;
+            rank += lower_rank(R, r2, e, r, eq);
+            /* an equal key at the insertion point is same-key and later in
+             * the order -> it supersedes e (newest-wins dedup) */
+            winner &= !eq;
+        }
+        uint64_t dlen = (uint64_t)e.full_size - 32 - e.klen;
+        bool keep = winner && (keep_tombstones || dlen != 0);
+        EntryMeta m;
+        m.src = ((uint64_t)r << 48) | e.off;
+        m.key_size = e.key_size;
+        m.full_size = e.full_size;
+        meta[rank] = m;
+        sizes[rank] = keep ? e.full_size : 0;
+        flags[rank] = keep ? 1u : 0u;
+    }
+}
+
+/* Output index records are the input format: offset u64 | key_size u32 |
+ * full_size u32 (entry_writer.rs:79-87, offsets recomputed from 0). */
+__global__ void k_emit(const EntryMeta* meta, const uint64_t* sizes,
+                       const uint64_t* dst_off, const uint32_t* pos,
+                       uint64_t total, uint8_t* out_index,
+                       uint64_t* src_map) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t g = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         g < total; g += stride) {
+        if (!sizes[g]) continue;
+        uint32_t p = pos[g];
+        uint8_t* rec = out_index + (uint64_t)p * 16;
+        uint64_t off = dst_off[g];
+        EntryMeta m = meta[g];
+        __builtin_memcpy(rec, &off, 8);
+        __builtin_memcpy(rec + 8, &m.key_size, 4);
+        __builtin_memcpy(rec + 12, &m.full_size, 4);
+        src_map[p] = m.src;
+    }
+}
+
+/* Balanced verbatim copy: 4-KiB destination window per 256-thread block,
+ * one 16-B granule per thread. A window can intersect at most
+ * 4096/32 + 2 entries (min entry size 32 B), so SPAN=136 offsets staged in
+ * LDS always cover it; a 16-B granule straddles at most one entry boundary.
+ */
+#define COPY_BLOCK 256
+#define COPY_WINDOW 4096
+#define COPY_SPAN 136
+
+__global__ __launch_bounds__(COPY_BLOCK) void k_copy(
+    RunsDesc R, const uint8_t* out_index, const uint64_t* src_map,
+    uint64_t n_surv, uint64_t total_bytes, uint8_t* out_data) {
+    __shared__ uint64_t s_off[COPY_SPAN + 1];
+    __shared__ uint64_t s_src[COPY_SPAN];
+    __shared__ uint64_t s_p0;
+
+    uint64_t n_windows = (total_bytes + COPY_WINDOW - 1) / COPY_WINDOW;
+    for (uint64_t w = blockIdx.x; w < n_windows; w += gridDim.x) {
+        uint64_t wstart = w * COPY_WINDOW;
+        /* largest p with offset(p) <= wstart */
+        if (threadIdx.x == 0) {
+            uint64_t lo = 0, hi = n_surv; /* find first > wstart, minus 1 */
+            while (lo < hi) {
+                uint64_t mid = (lo + hi) >> 1;
+                uint64_t o = ld_u64(out_index + mid * 16);
+                if (o <= wstart)
+                    lo = mid + 1;
+                else
+                    hi = mid;
+            }
+            s_p0 = lo - 1; /* offset(0)=0 <= wstart always, so lo >= 1 */
+        }
+        __syncthreads();
+        uint64_t p0 = s_p0;
+        uint32_t cnt = (uint32_t)((n_surv - p0) < COPY_SPAN ? (n_surv - p0)
+                                                            : COPY_SPAN);
+        if (threadIdx.x <= cnt && threadIdx.x < COPY_SPAN + 1) {
+            uint64_t p = p0 + threadIdx.x;
+            if (threadIdx.x == cnt)
+                s_off[cnt] = (p < n_surv) ? ld_u64(out_index + p * 16)
+                                          : total_bytes;
+            else {
+                s_off[threadIdx.x] = ld_u64(out_index + p * 16);
+                s_src[threadIdx.x] = src_map[p];
+            }
+        }
+        __syncthreads();
+
+        uint64_t gpos = wstart + (uint64_t)threadIdx.x * 16;
+        if (gpos < total_bytes) {
+            /* find j in [0, cnt): largest with s_off[j] <= gpos */
+            uint32_t lo = 0, hi = cnt;
+            while (lo < hi) {
+                uint32_t mid = (lo + hi) >> 1;
+                if (s_off[mid] <= gpos)
+                    lo = mid + 1;
+                else
+                    hi = mid;
+            }
+            uint32_t j = lo - 1;
+            uint64_t e_end = s_off[j + 1];
+            uint64_t src_pack = s_src[j];
+            const uint8_t* src = R.data[src_pack >> 48] +
+                                 (src_pack & 0xFFFFFFFFFFFFull) +
+                                 (gpos - s_off[j]);
+            uint32_t nbytes =
+                (uint32_t)((total_bytes - gpos) < 16 ? (total_bytes - gpos)
+                                                     : 16);
+            uint8_t* dst = out_data + gpos;
+            if (gpos + nbytes <= e_end) {
+                if (nbytes == 16) {
+                    uint4 v;
+                    __builtin_memcpy(&v, src, 16);
+                    *reinterpret_cast<uint4*>(dst) = v;
+                } else {
+                    for (uint32_t b = 0; b < nbytes; b++) dst[b] = src[b];
+                }
+            } else {
+                /* one entry boundary inside the granule */
+                uint32_t c1 = (uint32_t)(e_end - gpos);
+                for (uint32_t b = 0; b < c1; b++) dst[b] = src[b];
+                uint64_t src2_pack = s_src[j + 1];
+                const uint8_t* src2 = R.data[src2_pack >> 48] +
+                                      (src2_pack & 0xFFFFFFFFFFFFull);
+                for (uint32_t b = c1; b < nbytes; b++)
+                    dst[b] = src2[b - c1];
+            }
+        }
+        __syncthreads();
+    }
+}
+
+/* ------------------------------------------------------------------ */
+/* Host side                                                          */
+/* ------------------------------------------------------------------ */
+
+struct dbeel_gpu_job {
+    int device = -1;
+    hipStream_t stream = nullptr;
+    hipEvent_t ev[8] = {};
+    RunsDesc desc{};
+    uint8_t* d_input = nullptr; /* one slab: all run data+index            */
+    EntryMeta* d_meta = nullptr;
+    uint64_t* d_sizes = nullptr;
+    uint32_t* d_flags = nullptr;
+    uint64_t* d_dstoff = nullptr;
+    uint32_t* d_pos = nullptr;
+    uint8_t* d_outindex = nullptr;
+    uint64_t* d_srcmap = nullptr;
+    uint8_t* d_outdata = nullptr;
+    void* d_scantmp = nullptr;
+    size_t scantmp_bytes = 0;
+    uint32_t* d_err = nullptr;
+    uint64_t total_entries = 0;
+    uint64_t total_data_bytes = 0;
+    uint64_t input_bytes = 0;
+    /* last-run results */
+    uint64_t out_data_len = 0;
+    uint64_t out_entries = 0;
+    bool have_result = false;
+    double h2d_ms = 0.0;
+};
+
+static int validate_runs(const dbeel_run_view* runs, size_t n_runs) {
+    if (!runs || n_runs == 0) {
+        set_err("runs is null or empty");
+        return DBEEL_ERR_INVALID_ARG;
+    }
+    if (n_runs > MAX_RUNS) {
+        set_err("n_runs %zu exceeds MAX_RUNS %d", n_runs, MAX_RUNS);
+        return DBEEL_ERR_INVALID_ARG;
+    }
+    for (size_t r = 0; r < n_runs; r++) {
+        if (runs[r].index_len % 16) {
+            set_err("run %zu: index_len %zu not a multiple of 16", r,
+                    runs[r].index_len);
+            return DBEEL_ERR_CORRUPT;
+        }
+        if ((runs[r].data_len && !runs[r].data) ||
+            (runs[r].index_len && !runs[r].index)) {
+            set_err("run %zu: null pointer with nonzero length", r);
+            return DBEEL_ERR_INVALID_ARG;
+        }
+        if (runs[r].data_len >= (1ull << 48)) {
+            set_err("run %zu: data_len too large", r);
+            return DBEEL_ERR_ITEM_TOO_LARGE;
+        }
+    }
+    return DBEEL_OK;
+}
+
+extern "C" int dbeel_gpu_job_create(const dbeel_run_view* runs, size_t n_runs,
+                                    int device, dbeel_gpu_job** out_job) {
+    g_err[0] = 0;
+    if (!out_job) {
+        set_err("out_job is null");
+        return DBEEL_ERR_INVALID_ARG;
+    }
+    *out_job = nullptr;
+    int rc = validate_runs(runs, n_runs);
+    if (rc) return rc;
+    if (device < 0) {
+        set_err("device must be >= 0 (no CPU fallback in the product "
+                "library; the CPU restatement lives in oracle/liboracle.so "
+                "and is test infrastructure only)");
+        return DBEEL_ERR_INVALID_ARG;
+    }
+    int ndev = 0;
+    hipError_t de = hipGetDeviceCount(&ndev);
+    if (de != hipSuccess || device >= ndev) {
+        set_err("no usable HIP device %d (count=%d, %s)", device, ndev,
+                hipGetErrorString(de));
+        return DBEEL_ERR_NO_GPU;
+    }
+    HIP_CHECK(hipSetDevice(device));
+
+    dbeel_gpu_job* job = new (std::nothrow) dbeel_gpu_job();
+    if (!job) return DBEEL_ERR_OOM;
+    job->device = device;
+
+    uint64_t input_bytes = 0, total = 0, total_data = 0;
+    for (size_t r = 0; r < n_runs; r++) {
+        input_bytes += runs[r].data_len + runs[r].index_len;
+        total += runs[r].index_len / 16;
+        total_data += runs[r].data_len;
+    }
+    job->total_entries = total;
+    job->total_data_bytes = total_data;
+    job->input_bytes = input_bytes;
+
+#define JOB_CHECK(call)                                                     \
+    do {                                                                    \
+        hipError_t _e = (call);                                             \
+        if (_e != hipSuccess) {                                             \
+            set_err("%s failed: %s", #call, hipGetErrorString(_e));         \
+            dbeel_gpu_job_destroy(job);                                     \
+            return (_e == hipErrorOutOfMemory) ? DBEEL_ERR_OOM              \
+                                               : DBEEL_ERR_HIP;             \
+        }                                                                   \
+    } while (0)
+
+    JOB_CHECK(hipStreamCreate(&job->stream));
+    for (int i = 0; i < 8; i++) JOB_CHECK(hipEventCreate(&job->ev[i]));
+
+    JOB_CHECK(hipMalloc(&job->d_input, input_bytes ? input_bytes : 16));
+    uint64_t n = total ? total : 1;
+    JOB_CHECK(hipMalloc(&job->d_meta, n * sizeof(EntryMeta)));
+    JOB_CHECK(hipMalloc(&job->d_sizes, n * sizeof(uint64_t)));
+    JOB_CHECK(hipMalloc(&job->d_flags, n * sizeof(uint32_t)));
+    JOB_CHECK(hipMalloc(&job->d_dstoff, n * sizeof(uint64_t)));
+    JOB_CHECK(hipMalloc(&job->d_pos, n * sizeof(uint32_t)));
+    JOB_CHECK(hipMalloc(&job->d_outindex, n * 16));
+    JOB_CHECK(hipMalloc(&job->d_srcmap, n * sizeof(uint64_t)));
+    JOB_CHECK(hipMalloc(&job->d_outdata, total_data ? total_data : 16));
+    JOB_CHECK(hipMalloc(&job->d_err, 2 * sizeof(uint32_t)));
+
+    size_t t1 = 0, t2 = 0;
+    rocprim::exclusive_scan(nullptr, t1, job->d_sizes, job->d_dstoff,
+                            (uint64_t)0, n, rocprim::plus<uint64_t>(),
+                            job->stream);
+    rocprim::exclusive_scan(nullptr, t2, job->d_flags, job->d_pos,
+                            (uint32_t)0, n, rocprim::plus<uint32_t>(),
+                            job->stream);
+    job->scantmp_bytes = t1 > t2 ? t1 : t2;
+    JOB_CHECK(hipMalloc(&job->d_scantmp, job->scantmp_bytes));
+
+    /* Upload: one slab; record per-run device pointers. */
+    JOB_CHECK(hipEventRecord(job->ev[0], job->stream));
+    uint8_t* p = job->d_input;
+    RunsDesc& D = job->desc;
+    memset(&D, 0, sizeof D);
+    D.n_runs = (int)n_runs;
+    D.total = total;
+    uint64_t base = 0;
+    for (size_t r = 0; r < n_runs; r++) {
+        D.data[r] = p;
+        D.data_len[r] = runs[r].data_len;
+        if (runs[r].data_len) {
+            JOB_CHECK(hipMemcpyAsync(p, runs[r].data, runs[r].data_len,
+                                     hipMemcpyHostToDevice, job->stream));
+            p += runs[r].data_len;
+        }
+        D.index[r] = p;
+        D.count[r] = runs[r].index_len / 16;
+        if (runs[r].index_len) {
+            JOB_CHECK(hipMemcpyAsync(p, runs[r].index, runs[r].index_len,
+                                     hipMemcpyHostToDevice, job->stream));
+            p += runs[r].index_len;
+        }
+        D.entry_base[r] = base;
+        base += D.count[r];
+    }
+    JOB_CHECK(hipEventRecord(job->ev[1], job->stream));
+    JOB_CHECK(hipStreamSynchronize(job->stream));
+    float ms = 0;
+    JOB_CHECK(hipEventElapsedTime(&ms, job->ev[0], job->ev[1]));
+    job->h2d_ms = ms;
+
+    *out_job = job;
+    return DBEEL_OK;
+#undef JOB_CHECK
+}
+
+extern "C" void dbeel_gpu_job_destroy(dbeel_gpu_job* job) {
+    if (!job) return;
+    if (job->device >= 0) hipSetDevice(job->device);
+    hipFree(job->d_input);
+    hipFree(job->d_meta);
+    hipFree(job->d_sizes);
+    hipFree(job->d_flags);
+    hipFree(job->d_dstoff);
+    hipFree(job->d_pos);
+    hipFree(job->d_outindex);
+    hipFree(job->d_srcmap);
+    hipFree(job->d_outdata);
+    hipFree(job->d_scantmp);
+    hipFree(job->d_err);
+    for (int i = 0; i < 8; i++)
+        if (job->ev[i]) hipEventDestroy(job->ev[i]);
+    if (job->stream) hipStreamDestroy(job->stream);
+    delete job;
+}
+
+static uint32_t pick_grid(uint64_t work_items, uint32_t block) {
+    uint64_t blocks = (work_items + block - 1) / block;
+    /* 256 CUs x 8 workgroups/CU = 2048; cap and grid-stride the rest */
+    if (blocks > 2048) blocks = 2048;
+    if (blocks == 0) blocks = 1;
+    return (uint32_t)blocks;
+}
+
+extern "C" int dbeel_gpu_job_run(dbeel_gpu_job* job, int keep_tombstones,
+                                 uint64_t* out_data_len,
+                                 uint64_t* out_entries,
+                                 dbeel_compact_timings* t) {
+    g_err[0] = 0;
+    if (!job) {
+        set_err("job is null");
+        return DBEEL_ERR_INVALID_ARG;
+    }
+    HIP_CHECK(hipSetDevice(job->device));
+    hipStream_t s = job->stream;
+    uint64_t n = job->total_entries;
+    job->have_result = false;
+
+    HIP_CHECK(hipMemsetAsync(job->d_err, 0, 2 * sizeof(uint32_t), s));
+
+    HIP_CHECK(hipEventRecord(job->ev[0], s));
+    if (n) {
+        uint32_t grid = pick_grid(n, 256);
+        hipLaunchKernelGGL(k_validate, dim3(grid), dim3(256), 0, s, job->desc,
+                           job->d_err);
+        hipLaunchKernelGGL(k_rank, dim3(grid), dim3(256), 0, s, job->desc,
+                           job->d_meta, job->d_sizes, job->d_flags,
+                           keep_tombstones);
+    }
+    HIP_CHECK(hipEventRecord(job->ev[1], s));
+    if (n) {
+        size_t tmp = job->scantmp_bytes;
+        rocprim::exclusive_scan(job->d_scantmp, tmp, job->d_sizes,
+                                job->d_dstoff, (uint64_t)0, n,
+                                rocprim::plus<uint64_t>(), s);
+        tmp = job->scantmp_bytes;
+        rocprim::exclusive_scan(job->d_scantmp, tmp, job->d_flags, job->d_pos,
+                                (uint32_t)0, n, rocprim::plus<uint32_t>(), s);
+    }
+    HIP_CHECK(hipEventRecord(job->ev[2], s));
+    if (n) {
+        uint32_t grid = pick_grid(n, 256);
+        hipLaunchKernelGGL(k_emit, dim3(grid), dim3(256), 0, s, job->d_meta,
+                           job->d_sizes, job->d_dstoff, job->d_pos, n,
+                           job->d_outindex, job->d_srcmap);
+    }
+    HIP_CHECK(hipEventRecord(job->ev[3], s));
+
+    /* Need totals on host to size/launch the copy; one small sync. */
+    uint64_t last_size = 0, last_off = 0;
+    uint32_t last_flag = 0, last_pos = 0, err = 0;
+    if (n) {
+        HIP_CHECK(hipMemcpyAsync(&last_size, job->d_sizes + (n - 1), 8,
+                                 hipMemcpyDeviceToHost, s));
+        HIP_CHECK(hipMemcpyAsync(&last_off, job->d_dstoff + (n - 1), 8,
+                                 hipMemcpyDeviceToHost, s));
+        HIP_CHECK(hipMemcpyAsync(&last_flag, job->d_flags + (n - 1), 4,
+                                 hipMemcpyDeviceToHost, s));
+        HIP_CHECK(hipMemcpyAsync(&last_pos, job->d_pos + (n - 1), 4,
+                                 hipMemcpyDeviceToHost, s));
+    }
+    HIP_CHECK(hipMemcpyAsync(&err, job->d_err, 4, hipMemcpyDeviceToHost, s));
+    HIP_CHECK(hipStreamSynchronize(s));
+    if (err) {
+        set_err(err & DERR_UNSORTED
+                    ? "input run not strictly sorted by unique keys "
+                      "(dbeel flush invariant violated)"
+                    : "corrupt entry/index record");
+        return DBEEL_ERR_CORRUPT;
+    }
+    uint64_t total_out = last_off + last_size;
+    uint64_t n_surv = (uint64_t)last_pos + last_flag;
+
+    HIP_CHECK(hipEventRecord(job->ev[4], s));
+    if (total_out) {
+        uint64_t windows = (total_out + COPY_WINDOW - 1) / COPY_WINDOW;
+        uint32_t grid = windows > 4096 ? 4096 : (uint32_t)windows;
+        hipLaunchKernelGGL(k_copy, dim3(grid), dim3(COPY_BLOCK), 0, s,
+                           job->desc, job->d_outindex, job->d_srcmap, n_surv,
+                           total_out, job->d_outdata);
+    }
+    HIP_CHECK(hipEventRecord(job->ev[5], s));
+    HIP_CHECK(hipStreamSynchronize(s));
+    HIP_CHECK(hipGetLastError());
+
+    job->out_data_len = total_out;
+    job->out_entries = n_surv;
+    job->have_result = true;
+    if (out_data_len) *out_data_len = total_out;
+    if (out_entries) *out_entries = n_surv;
+    if (t) {
+        float rank_ms = 0, scan_ms = 0, emit_ms = 0, copy_ms = 0;
+        hipEventElapsedTime(&rank_ms, job->ev[0], job->ev[1]);
+        hipEventElapsedTime(&scan_ms, job->ev[1], job->ev[2]);
+        hipEventElapsedTime(&emit_ms, job->ev[2], job->ev[3]);
+        hipEventElapsedTime(&copy_ms, job->ev[4], job->ev[5]);
+        t->h2d_ms = job->h2d_ms;
+        t->rank_ms = rank_ms;
+        t->scan_ms = scan_ms;
+        t->emit_ms = emit_ms;
+        t->copy_ms = copy_ms;
+        t->kernel_ms = rank_ms + scan_ms + emit_ms + copy_ms;
+        t->d2h_ms = 0.0;
+    }
+    return DBEEL_OK;
+}
+
+extern "C" int dbeel_gpu_job_fetch(dbeel_gpu_job* job,
+                                   dbeel_compact_result* out) {
+    g_err[0] = 0;
+    if (!job || !out || !job->have_result) {
+        set_err("job_fetch: no result available");
+        return DBEEL_ERR_INVALID_ARG;
+    }
+    HIP_CHECK(hipSetDevice(job->device));
+    memset(out, 0, sizeof *out);
+    uint64_t dlen = job->out_data_len, ilen = job->out_entries * 16;
+    out->data = (uint8_t*)malloc(dlen ? dlen : 1);
+    out->index = (uint8_t*)malloc(ilen ? ilen : 1);
+    if (!out->data || !out->index) {
+        free(out->data);
+        free(out->index);
+        memset(out, 0, sizeof *out);
+        set_err("host malloc failed");
+        return DBEEL_ERR_OOM;
+    }
+    if (dlen)
+        HIP_CHECK(hipMemcpyAsync(out->data, job->d_outdata, dlen,
+                                 hipMemcpyDeviceToHost, job->stream));
+    if (ilen)
+        HIP_CHECK(hipMemcpyAsync(out->index, job->d_outindex, ilen,
+                                 hipMemcpyDeviceToHost, job->stream));
+    HIP_CHECK(hipStreamSynchronize(job->stream));
+    out->data_len = dlen;
+    out->index_len = ilen;
+    out->entries_written = job->out_entries;
+    return DBEEL_OK;
+}
+
+extern "C" int dbeel_gpu_compact_timed(const dbeel_run_view* runs,
+                                       size_t n_runs, int keep_tombstones,
+                                       int device, dbeel_compact_result* out,
+                                       dbeel_compact_timings* t) {
+    if (!out) {
+        set_err("out is null");
+        return DBEEL_ERR_INVALID_ARG;
+    }
+    dbeel_gpu_job* job = nullptr;
+    int rc = dbeel_gpu_job_create(runs, n_runs, device, &job);
+    if (rc) return rc;
+    rc = dbeel_gpu_job_run(job, keep_tombstones, nullptr, nullptr, t);
+    if (!rc) rc = dbeel_gpu_job_fetch(job, out);
+    dbeel_gpu_job_destroy(job);
+    return rc;
+}
+
+extern "C" int dbeel_gpu_compact(const dbeel_run_view* runs, size_t n_runs,
+                                 int keep_tombstones, int device,
+                                 dbeel_compact_result* out) {
+    return dbeel_gpu_compact_timed(runs, n_runs, keep_tombstones, device, out,
+                                   nullptr);
+}
+
+extern "C" void dbeel_gpu_result_free(dbeel_compact_result* r) {
+    if (!r) return;
+    free(r->data);
+    free(r->index);
+    memset(r, 0, sizeof *r);
+}

@@ -31,6 +31,7 @@ def load_golden(name):
 
 GOLDEN_CASES = [
     "basic",
+    "ts_shuffle",
     "tie_ts",
     "all_tombstones",
     "empty_run",

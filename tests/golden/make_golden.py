@@ -137,8 +137,25 @@ def case_disjoint(rng):
     ]
 
 
+def case_ts_shuffle(rng):
+    # Winner is max (timestamp, run index), NOT the highest run: give some
+    # keys their newest timestamp in an OLDER run.
+    keys = [bytes([i, 7]) for i in range(30)]
+    runs = []
+    ts = rng.permutation(90).reshape(3, 30)
+    for r in range(3):
+        runs.append(
+            sorted_run(
+                [e(k, bytes([r]) + k, int(ts[r][i]) - 40)
+                 for i, k in enumerate(keys)]
+            )
+        )
+    return runs
+
+
 CASES = {
     "basic": case_basic,
+    "ts_shuffle": case_ts_shuffle,
     "tie_ts": case_tie_ts,
     "all_tombstones": case_all_tombstones,
     "empty_run": case_empty_run,

@@ -1,0 +1,83 @@
+"""CPU-side ABI checks: the product library builds for gfx950, loads, and
+exports every symbol include/dbeel_gpu.h declares. No compute without a GPU.
+"""
+import ctypes
+import os
+
+import pytest
+
+from conftest import REPO_ROOT
+
+LIB = os.path.join(REPO_ROOT, "dbeel_amd", "libdbeel_gpu.so")
+
+HEADER_SYMBOLS = [
+    "dbeel_gpu_compact",
+    "dbeel_gpu_compact_timed",
+    "dbeel_gpu_result_free",
+    "dbeel_gpu_last_error",
+    "dbeel_gpu_job_create",
+    "dbeel_gpu_job_run",
+    "dbeel_gpu_job_fetch",
+    "dbeel_gpu_job_destroy",
+]
+
+
+def _built():
+    if not os.path.exists(LIB):
+        import __graft_entry__
+
+        __graft_entry__.build()
+    return LIB
+
+
+def test_product_lib_exports_header_symbols():
+    lib = ctypes.CDLL(_built())
+    for sym in HEADER_SYMBOLS:
+        assert getattr(lib, sym, None) is not None, sym
+
+
+def test_header_declares_every_symbol():
+    hdr = open(os.path.join(REPO_ROOT, "include", "dbeel_gpu.h")).read()
+    for sym in HEADER_SYMBOLS:
+        assert sym in hdr, sym
+
+
+def test_device_minus_one_rejected():
+    """device=-1 must NOT fall back to a CPU path (product/oracle
+    separation); it is an invalid argument by design."""
+    _built()
+    import dbeel_amd
+    from dbeel_amd.engine import DbeelGpuError
+    from dbeel_amd.format import Entry, build_run
+
+    runs = [build_run([Entry(b"a", b"b", 1)])]
+    with pytest.raises(DbeelGpuError) as ei:
+        dbeel_amd.compact(runs, keep_tombstones=True, device=-1)
+    assert ei.value.code == 1  # INVALID_ARG
+
+
+def test_no_gpu_errors_loudly():
+    """On a GPU-less host, using a device ordinal must raise, never silently
+    fall back."""
+    _built()
+    import dbeel_amd
+    from dbeel_amd.engine import DbeelGpuError
+    from dbeel_amd.format import Entry, build_run
+
+    runs = [build_run([Entry(b"a", b"b", 1)])]
+    try:
+        dbeel_amd.compact(runs, keep_tombstones=True, device=0)
+    except DbeelGpuError as e:
+        assert e.code in (4, 5)  # HIP / NO_GPU
+    else:
+        # a GPU is actually present (gpurun box) — fine
+        pass
+
+
+def test_oracle_lib_exports():
+    import oracle
+
+    lib = ctypes.CDLL(oracle.build())
+    for sym in ("dbeel_oracle_compact", "dbeel_oracle_result_free",
+                "dbeel_oracle_last_error"):
+        assert getattr(lib, sym, None) is not None, sym

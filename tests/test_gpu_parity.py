@@ -1,0 +1,142 @@
+"""GPU parity tests (marked gpu; run on a real MI355X via gpurun).
+
+The bar (SURVEY.md §8c / north_star): output `.data`/`.index` bytes from the
+HIP engine are BIT-IDENTICAL to the CPU oracle on the same inputs — golden
+fixtures, seeded BASELINE shapes, and edge cases. Plus size-independent
+property checks at larger scale.
+"""
+import numpy as np
+import pytest
+
+import oracle
+from conftest import GOLDEN_CASES, load_golden
+from dbeel_amd.format import Entry, build_run, parse_run
+from dbeel_amd.genruns import make_runs
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(scope="module")
+def engine():
+    import dbeel_amd
+
+    return dbeel_amd
+
+
+@pytest.mark.parametrize("name", GOLDEN_CASES)
+@pytest.mark.parametrize("keep", [True, False])
+def test_golden_parity(engine, name, keep):
+    runs, exp_keep, exp_drop = load_golden(name)
+    exp = exp_keep if keep else exp_drop
+    data, index, n = engine.compact(runs, keep_tombstones=keep, device=0)
+    assert index == exp[1]
+    assert data == exp[0]
+    assert n == len(exp[1]) // 16
+
+
+@pytest.mark.parametrize(
+    "shape",
+    [
+        # (n_runs, entries, ksize, vsize, overlap, tomb)
+        (2, 10_000, 16, 64, 0.0, 0.0),       # cfg1 exact
+        (4, 50_000, 16, 256, 0.3, 0.0),      # cfg2 scaled
+        (8, 30_000, 32, 1024, 0.5, 0.05),    # cfg3 scaled
+        (16, 8_000, 16, 256, 0.5, 0.2),      # 16-way merge
+        (1, 5_000, 16, 64, 0.0, 0.3),        # single run
+        (3, 2_000, 8, 32, 1.0, 0.5),         # full overlap, many tombstones
+    ],
+)
+@pytest.mark.parametrize("keep", [True, False])
+def test_seeded_shape_parity(engine, shape, keep):
+    n_runs, n, ks, vs, ov, tb = shape
+    runs = make_runs(n_runs, n, ks, vs, overlap_frac=ov, tombstone_frac=tb,
+                     seed=0xDBEE1 + n_runs)
+    gd, gi, gn = engine.compact(runs, keep_tombstones=keep, device=0)
+    od, oi, on = oracle.compact(runs, keep_tombstones=keep)
+    assert gn == on
+    assert gi == oi
+    assert gd == od
+
+
+def test_ragged_keys_parity(engine):
+    """Variable-length keys 1..128 B incl. prefix pairs and empty keys,
+    mixed value sizes (cfg5 precursor)."""
+    rng = np.random.default_rng(99)
+    runs = []
+    for r in range(6):
+        keys = {b"", b"\x00", b"\x00\x00"}
+        for _ in range(800):
+            keys.add(bytes(rng.integers(0, 256, int(rng.integers(1, 129)),
+                                        dtype=np.uint8)))
+        ents = []
+        for i, k in enumerate(sorted(keys)):
+            dlen = int(rng.integers(0, 6)) * 111
+            ents.append(Entry(k, bytes(rng.integers(0, 256, dlen,
+                                                    dtype=np.uint8)),
+                              int(rng.integers(-100, 100))))
+        runs.append(build_run(ents))
+    for keep in (True, False):
+        gd, gi, gn = engine.compact(runs, keep_tombstones=keep, device=0)
+        od, oi, on = oracle.compact(runs, keep_tombstones=keep)
+        assert (gd, gi, gn) == (od, oi, on)
+
+
+def test_empty_inputs(engine):
+    runs = [(b"", b""), (b"", b"")]
+    gd, gi, gn = engine.compact(runs, keep_tombstones=True, device=0)
+    assert (gd, gi, gn) == (b"", b"", 0)
+
+
+def test_corrupt_input_rejected(engine):
+    import struct
+
+    from dbeel_amd.engine import DbeelGpuError
+
+    # unsorted run (flush invariant violated) must error loudly
+    r1 = build_run([Entry(b"b", b"1", 1)])
+    r2 = build_run([Entry(b"a", b"2", 2)])
+    bad_data = r1[0] + r2[0]
+    bad_index = r1[1] + struct.pack("<QII", len(r1[0]), 9, 33)
+    with pytest.raises(DbeelGpuError) as ei:
+        engine.compact([(bad_data, bad_index)], keep_tombstones=True, device=0)
+    assert ei.value.code == 2  # CORRUPT
+
+    # index record pointing past data_len
+    idx = struct.pack("<QII", 1000, 9, 64)
+    with pytest.raises(DbeelGpuError) as ei:
+        engine.compact([(b"\x00" * 64, idx)], keep_tombstones=True, device=0)
+    assert ei.value.code == 2
+
+
+def test_large_scale_properties(engine):
+    """Mid-scale run (~600 MB input) — parity vs oracle would be slow at
+    full BASELINE size, so check parity at this size and properties that are
+    size-independent: sortedness, unique keys, verbatim bytes, count."""
+    runs = make_runs(8, 120_000, 32, 1024, overlap_frac=0.5,
+                     tombstone_frac=0.05, seed=0xC0FFEE)
+    gd, gi, gn = engine.compact(runs, keep_tombstones=False, device=0)
+    od, oi, on = oracle.compact(runs, keep_tombstones=False)
+    assert gn == on and gi == oi and gd == od
+
+    ents = parse_run(gd, gi)
+    keys = [e.key for e in ents]
+    assert keys == sorted(keys)
+    assert len(set(keys)) == len(keys)
+    assert not any(e.is_tombstone for e in ents)
+
+
+def test_resident_job_repeatable(engine):
+    """Job API: repeated runs on resident inputs give identical results and
+    both keep_tombstones settings work on one upload."""
+    runs = make_runs(4, 20_000, 16, 256, overlap_frac=0.4,
+                     tombstone_frac=0.1, seed=5)
+    with engine.Job(runs, device=0) as job:
+        d1, n1, t1 = job.run(keep_tombstones=False)
+        data1, index1, _ = job.fetch()
+        d2, n2, t2 = job.run(keep_tombstones=False)
+        data2, index2, _ = job.fetch()
+        assert (d1, n1) == (d2, n2)
+        assert data1 == data2 and index1 == index2
+        assert t2["kernel_ms"] > 0
+        od, oi, on = oracle.compact(runs, keep_tombstones=False)
+        assert (data1, index1, n1) == (od, oi, on)

@@ -76,7 +76,9 @@ def test_oracle_agrees_with_pymerge_random():
             for i, k in enumerate(keys):
                 dlen = int(rng.integers(0, 4)) * 33
                 data = bytes(rng.integers(0, 256, dlen, dtype=np.uint8))
-                ents.append(Entry(k, data, (r << 20) + i))
+                # random timestamps (collisions across runs exercise the
+                # run-index tie-break; newest-in-older-run also covered)
+                ents.append(Entry(k, data, int(rng.integers(-50, 50))))
             runs.append(build_run(ents))
         for keep in (True, False):
             exp = pymerge_merge(runs, keep)
