@@ -1,0 +1,229 @@
+#!/usr/bin/env python3
+"""Benchmark: dbeel SSTable compaction on MI355X.
+
+Workload (BASELINE.json metric config — the 8-run merge the headline metric
+is quoted on): config 3 = 8 runs x ~1 GiB, 32 B keys / 1 KiB values, 50% key
+overlap, 5% tombstones, synthetic seeded runs (no network). A "step" = one
+full compaction of the 8-run set with inputs already resident in HBM
+(outputs land in HBM; PCIe-inclusive rates are reported in DESIGN.md, never
+as `value`).
+
+Multi-GPU: the path shards as INDEPENDENT jobs — one
+compaction job per GPU (weak scaling), with the only collective an RCCL
+all-gather of per-rank emitted byte counts (8 x u64-scale) per step, per the
+north star.
+
+Usage:  python bench.py [--gpus N] [--steps K] [--warmup W] [--scale S]
+For N>1 the driver launches via torch.distributed.run (one rank per GPU);
+rank/device from RANK/LOCAL_RANK/WORLD_SIZE.
+"""
+import argparse
+import json
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+HBM_PEAK_GBPS = 8000.0  # MI355X HBM3E spec peak, GB/s (MI355X_MICROARCH.md)
+
+
+def make_workload(rank: int, scale: float):
+    from dbeel_amd.genruns import CONFIGS, make_runs
+
+    cfg = dict(CONFIGS["cfg3"])
+    if scale != 1.0:
+        cfg["entries_per_run"] = max(64, int(cfg["entries_per_run"] * scale))
+    runs = make_runs(seed=0xDBEE1 + 7919 * rank, **cfg)
+    return runs, cfg
+
+
+def rank_algorithmic_bytes(cfg, n_entries: int) -> int:
+    # k_rank minimum traffic: each entry's index record (16 B) + key +
+    # trailing timestamp (16 B) read once.
+    return n_entries * (16 + cfg["key_size"] + 16)
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=10)
+    ap.add_argument("--warmup", type=int, default=3)
+    ap.add_argument("--scale", type=float, default=1.0,
+                    help="scale entries_per_run (1.0 = full config 3)")
+    ap.add_argument("--keep-tombstones", action="store_true")
+    ap.add_argument("--cpu-baseline-scale", type=float, default=0.25,
+                    help="fraction of the workload timed on 1 host core")
+    ap.add_argument("--skip-cpu-baseline", action="store_true")
+    args = ap.parse_args()
+
+    import torch
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    n_gpus = max(world, args.gpus)
+    dist = None
+    if world > 1:
+        import torch.distributed as dist_mod
+
+        dist = dist_mod
+        dist.init_process_group(backend="nccl")
+        torch.cuda.set_device(local_rank)
+
+    import dbeel_amd
+
+    keep = bool(args.keep_tombstones)
+    runs, cfg = make_workload(rank, args.scale)
+    input_bytes = sum(d.nbytes + i.nbytes for d, i in runs)
+    n_entries = sum(i.nbytes // 16 for d, i in runs)
+
+    job = dbeel_amd.Job(runs, device=local_rank)
+
+    # Warmup (untimed)
+    out_bytes = out_entries = 0
+    tim_acc = None
+    for _ in range(args.warmup):
+        out_bytes, out_entries, _ = job.run(keep)
+
+    def barrier_sync():
+        torch.cuda.synchronize(local_rank) if torch.cuda.is_available() else None
+        if dist:
+            dist.barrier()
+        if torch.cuda.is_available():
+            torch.cuda.synchronize(local_rank)
+
+    barrier_sync()
+    t0 = time.perf_counter()
+    tims = []
+    for _ in range(args.steps):
+        out_bytes, out_entries, t = job.run(keep)
+        tims.append(t)
+        if dist:
+            # the path's only collective: RCCL all-gather of emitted byte
+            # counts over xGMI (north_star / SURVEY.md §5)
+            counts = torch.tensor([out_bytes], dtype=torch.int64,
+                                  device=f"cuda:{local_rank}")
+            gathered = [torch.zeros_like(counts) for _ in range(world)]
+            dist.all_gather(gathered, counts)
+    barrier_sync()
+    elapsed = time.perf_counter() - t0
+
+    # MAX over ranks
+    if dist:
+        e = torch.tensor([elapsed], dtype=torch.float64,
+                         device=f"cuda:{local_rank}")
+        dist.all_reduce(e, op=dist.ReduceOp.MAX)
+        elapsed = float(e.item())
+
+    ms_per_step = elapsed * 1000.0 / args.steps
+    total_input = input_bytes * n_gpus
+    value = (total_input / 1e6) / (elapsed / args.steps)  # MB/s, whole job
+
+    avg = {k: sum(t[k] for t in tims) / len(tims) for k in tims[0]}
+    # dominant kernel by average time
+    per_kernel = {
+        "prep": avg["prep_ms"],
+        "rank": avg["rank_ms"],
+        "scan": avg["scan_ms"],
+        "emit": avg["emit_ms"],
+        "copy": avg["copy_ms"],
+    }
+    dom = max(per_kernel, key=per_kernel.get)
+    out_index_bytes = out_entries * 16
+    algo = {
+        # per-kernel algorithmic bytes
+        "prep": n_entries * (16 + cfg["key_size"] + 16 + 8),
+        "rank": rank_algorithmic_bytes(cfg, n_entries),
+        "scan": 3 * 8 * n_entries + 2 * 4 * n_entries,
+        "emit": 16 * n_entries + 24 * out_entries,
+        "copy": 2 * out_bytes + 24 * out_entries,
+    }
+    dom_ms = per_kernel[dom]
+    achieved = (algo[dom] / 1e9) / (dom_ms / 1e3) if dom_ms > 0 else 0.0
+    roofline = {
+        "bound": "hbm",
+        "kernel": f"k_{dom}",
+        "achieved": round(achieved, 1),
+        "peak": HBM_PEAK_GBPS,
+        "unit": "GB/s",
+        "frac": round(achieved / HBM_PEAK_GBPS, 4),
+        "traffic": None,  # PMC-measured per-launch HBM bytes: profiles/
+    }
+    # whole-pipeline roofline (SURVEY.md §8d algorithmic B / kernel time)
+    B = input_bytes + out_bytes + out_index_bytes
+    pipe_gbps = (B / 1e9) / (avg["kernel_ms"] / 1e3)
+    roofline_pipeline = {
+        "bound": "hbm",
+        "achieved": round(pipe_gbps, 1),
+        "peak": HBM_PEAK_GBPS,
+        "unit": "GB/s",
+        "frac": round(pipe_gbps / HBM_PEAK_GBPS, 4),
+        "traffic": None,
+    }
+
+    cpu_baseline = None
+    if rank == 0 and n_gpus == 1 and not args.skip_cpu_baseline:
+        import oracle
+
+        bs = args.cpu_baseline_scale * args.scale
+        cruns, _ = make_workload(0, bs)
+        cbytes = sum(d.nbytes + i.nbytes for d, i in cruns)
+        c0 = time.perf_counter()
+        oracle.compact(cruns, keep_tombstones=keep)
+        c1 = time.perf_counter()
+        cpu_baseline = {
+            "value": round((cbytes / 1e6) / (c1 - c0), 1),
+            "unit": "MB/s",
+            "cores": 1,
+            "kind": "port",
+            "sample": (
+                f"cfg3 shape at {bs:.2f} scale "
+                f"({cbytes / 1e6:.0f} MB input, {c1 - c0:.1f}s on 1 core; "
+                "C oracle restatement — Rust/glommio unbuildable here, "
+                "BASELINE.md)"
+            ),
+        }
+
+    if rank == 0:
+        line = {
+            "metric": "compaction_MBps_input",
+            "value": round(value, 1),
+            "unit": "MB/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,  # no published compaction number
+            "dtype": "u8",
+            "data": "synthetic",
+            "config": {
+                "workload": "cfg3_8run_x_1GiB"
+                + (f"_scale{args.scale}" if args.scale != 1.0 else ""),
+                "n_runs": cfg["n_runs"],
+                "entries_per_run": cfg["entries_per_run"],
+                "key_size": cfg["key_size"],
+                "value_size": cfg["value_size"],
+                "overlap": cfg.get("overlap_frac", 0),
+                "tombstones": cfg.get("tombstone_frac", 0),
+                "keep_tombstones": keep,
+                "parallelism": f"independent_jobs_x{n_gpus}",
+            },
+            "kernel_ms": {k: round(v, 3) for k, v in avg.items()},
+            "out_bytes": out_bytes,
+            "out_entries": out_entries,
+            "roofline": roofline,
+            "roofline_pipeline": roofline_pipeline,
+            "cpu_baseline": cpu_baseline,
+        }
+        print(json.dumps(line))
+
+    job.close()
+    if dist:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -74,9 +74,17 @@ struct RunsDesc {
     uint64_t data_len[MAX_RUNS];
     uint64_t count[MAX_RUNS];       /* entries per run                  */
     uint64_t entry_base[MAX_RUNS];  /* exclusive prefix sum of count    */
+    uint32_t piv_base[MAX_RUNS];    /* exclusive prefix sum of n pivots */
     int n_runs;
     uint64_t total;                 /* sum of count                     */
+    uint32_t n_pivots;              /* total pivots over all runs       */
 };
+
+/* Pivot stride: crossranks of every PIV_STRIDE-th entry of each run into
+ * every other run are precomputed (k_pivots), bounding each entry's
+ * binary search to a ~PIV_STRIDE-wide, L2-resident window. */
+#define PIV_SHIFT 10
+#define PIV_STRIDE (1u << PIV_SHIFT)
 
 /* 16 B/entry scratch record, scattered by global rank in k_rank. */
 struct EntryMeta {
@@ -181,7 +189,26 @@ __device__ __forceinline__ int cmp_full(const EView& a, int ra,
 #define DERR_CORRUPT 1u
 #define DERR_UNSORTED 2u
 
-__global__ void k_validate(RunsDesc R, uint32_t* err) {
+/* Big-endian key prefix: the first min(8, klen) key bytes, zero-padded,
+ * byteswapped so u64 comparison == lexicographic byte comparison of those
+ * bytes. pfx(a) != pfx(b) implies sign(u64 cmp) == sign(key cmp); equality
+ * requires the full comparator (length / tail bytes / timestamp). */
+__device__ __forceinline__ uint64_t key_prefix(const uint8_t* key,
+                                               uint64_t klen) {
+    uint64_t v = 0;
+    if (klen >= 8) {
+        v = ld_u64(key);
+    } else {
+        for (uint64_t j = 0; j < klen; j++) v |= (uint64_t)key[j] << (8 * j);
+    }
+    return __builtin_bswap64(v);
+}
+
+/* Validates every entry (bounds + bincode field cross-check,
+ * read_next_entry lsm_tree.rs:1158-70), checks each run is strictly sorted
+ * by key (flush invariant), and extracts the dense key-prefix array the
+ * rank searches run on. */
+__global__ void k_prepare(RunsDesc R, uint64_t* pfx, uint32_t* err) {
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
     for (uint64_t g = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
          g < R.total; g += stride) {
@@ -192,9 +219,11 @@ __global__ void k_validate(RunsDesc R, uint32_t* err) {
         EView e;
         if (!load_entry(R, r, i, e)) {
             atomicOr(err, DERR_CORRUPT);
+            pfx[g] = 0;
             continue;
         }
-        /* bincode field cross-check (read_next_entry, lsm_tree.rs:1158-70) */
+        pfx[g] = key_prefix(e.key, e.klen);
+        /* bincode field cross-check */
         if (ld_u64(e.raw) != e.klen ||
             ld_u64(e.raw + 8 + e.klen) != (uint64_t)e.full_size - 32 - e.klen) {
             atomicOr(err, DERR_CORRUPT);
@@ -213,16 +242,29 @@ __global__ void k_validate(RunsDesc R, uint32_t* err) {
     }
 }
 
-/* Number of entries in run r2 strictly below e (full order). Runs are
- * sorted by key with unique keys, hence sorted in full order too. */
-__device__ __forceinline__ uint64_t lower_rank(const RunsDesc& R, int r2,
-                                               const EView& e, int re,
+/* Number of entries in run r2 (within [lo, hi)) strictly below e (full
+ * order). Runs are sorted by key with unique keys, hence sorted in full
+ * order too. Search steps compare dense 8-byte prefixes first; the full
+ * comparator (scattered entry read) runs only on prefix ties. */
+__device__ __forceinline__ uint64_t lower_rank(const RunsDesc& R,
+                                               const uint64_t* pfx, int r2,
+                                               const EView& e, uint64_t epfx,
+                                               int re, uint64_t lo,
+                                               uint64_t hi,
                                                bool& equal_key_at) {
-    uint64_t lo = 0, hi = R.count[r2];
+    const uint64_t* p2 = pfx + R.entry_base[r2];
     while (lo < hi) {
         uint64_t mid = (lo + hi) >> 1;
+        uint64_t pm = p2[mid];
+        if (pm != epfx) {
+            if (pm < epfx)
+                lo = mid + 1;
+            else
+                hi = mid;
+            continue;
+        }
         EView m;
-        /* corrupt records are flagged by k_validate; avoid the OOB read
+        /* corrupt records are flagged by k_prepare; avoid the OOB read
          * here (results are discarded once the host sees the flag) */
         if (!load_entry(R, r2, mid, m)) {
             hi = mid;
@@ -234,7 +276,7 @@ __device__ __forceinline__ uint64_t lower_rank(const RunsDesc& R, int r2,
             hi = mid;
     }
     equal_key_at = false;
-    if (lo < R.count[r2]) {
+    if (lo < R.count[r2] && p2[lo] == epfx) {
         EView m;
         if (load_entry(R, r2, lo, m))
             equal_key_at = (m.klen == e.klen) &&
@@ -243,8 +285,37 @@ __device__ __forceinline__ uint64_t lower_rank(const RunsDesc& R, int r2,
     return lo;
 }
 
-__global__ void k_rank(RunsDesc R, EntryMeta* meta, uint64_t* sizes,
-                       uint32_t* flags, int keep_tombstones) {
+/* Crossranks of every PIV_STRIDE-th entry of each run into every other
+ * run: piv[(piv_base[r] + j) * n_runs + r2] = lower_rank of entry
+ * (r, j*PIV_STRIDE) in run r2. Tiny kernel (N/1024 x n_runs searches). */
+__global__ void k_pivots(RunsDesc R, const uint64_t* pfx, uint32_t* piv) {
+    uint32_t stride = gridDim.x * blockDim.x;
+    for (uint32_t t = blockIdx.x * blockDim.x + threadIdx.x; t < R.n_pivots;
+         t += stride) {
+        int r = 0;
+        while (r + 1 < R.n_runs && t >= R.piv_base[r + 1]) r++;
+        uint64_t i = (uint64_t)(t - R.piv_base[r]) << PIV_SHIFT;
+        EView e;
+        if (!load_entry(R, r, i, e)) {
+            for (int r2 = 0; r2 < R.n_runs; r2++)
+                piv[(uint64_t)t * R.n_runs + r2] = 0;
+            continue;
+        }
+        uint64_t epfx = key_prefix(e.key, e.klen);
+        for (int r2 = 0; r2 < R.n_runs; r2++) {
+            uint64_t cr = 0;
+            if (r2 != r) {
+                bool eq;
+                cr = lower_rank(R, pfx, r2, e, epfx, r, 0, R.count[r2], eq);
+            }
+            piv[(uint64_t)t * R.n_runs + r2] = (uint32_t)cr;
+        }
+    }
+}
+
+__global__ void k_rank(RunsDesc R, const uint64_t* pfx, const uint32_t* piv,
+                       EntryMeta* meta, uint64_t* sizes, uint32_t* flags,
+                       int keep_tombstones) {
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
     for (uint64_t g = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
          g < R.total; g += stride) {
@@ -253,20 +324,33 @@ __global__ void k_rank(RunsDesc R, EntryMeta* meta, uint64_t* sizes,
         uint64_t i = g - R.entry_base[r];
         EView e;
         if (!load_entry(R, r, i, e)) {
-            /* k_validate has flagged this input; keep memory safe and park
+            /* k_prepare has flagged this input; keep memory safe and park
              * the entry at its local slot (results will be discarded) */
             meta[g] = EntryMeta{0, 8, 32};
             sizes[g] = 0;
             flags[g] = 0;
             continue;
         }
+        uint64_t epfx = pfx[g];
+
+        /* pivot window of this entry: search bounds per other run */
+        uint32_t j = (uint32_t)(i >> PIV_SHIFT);
+        uint32_t n_piv_r =
+            (uint32_t)((R.count[r] + PIV_STRIDE - 1) >> PIV_SHIFT);
+        const uint32_t* lo_cr = piv + ((uint64_t)R.piv_base[r] + j) * R.n_runs;
+        const uint32_t* hi_cr =
+            (j + 1 < n_piv_r)
+                ? piv + ((uint64_t)R.piv_base[r] + j + 1) * R.n_runs
+                : nullptr;
 
         uint64_t rank = i;
         bool winner = true;
         for (int r2 = 0; r2 < R.n_runs; r2++) {
             if (r2 == r) continue;
+            uint64_t lo = lo_cr[r2];
+            uint64_t hi = hi_cr ? (uint64_t)hi_cr[r2] : R.count[r2];
             bool eq;
-            rank += lower_rank(R, r2, e, r, eq);
+            rank += lower_rank(R, pfx, r2, e, epfx, r, lo, hi, eq);
             /* an equal key at the insertion point is same-key and later in
              * the order -> it supersedes e (newest-wins dedup) */
             winner &= !eq;
@@ -284,11 +368,12 @@ __global__ void k_rank(RunsDesc R, EntryMeta* meta, uint64_t* sizes,
 }
 
 /* Output index records are the input format: offset u64 | key_size u32 |
- * full_size u32 (entry_writer.rs:79-87, offsets recomputed from 0). */
-__global__ void k_emit(const EntryMeta* meta, const uint64_t* sizes,
-                       const uint64_t* dst_off, const uint32_t* pos,
-                       uint64_t total, uint8_t* out_index,
-                       uint64_t* src_map) {
+ * full_size u32 (entry_writer.rs:79-87, offsets recomputed from 0).
+ * src_map gets the ABSOLUTE device address of each survivor's bytes. */
+__global__ void k_emit(RunsDesc R, const EntryMeta* meta,
+                       const uint64_t* sizes, const uint64_t* dst_off,
+                       const uint32_t* pos, uint64_t total,
+                       uint8_t* out_index, uint64_t* src_map) {
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
     for (uint64_t g = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
          g < total; g += stride) {
@@ -300,74 +385,88 @@ __global__ void k_emit(const EntryMeta* meta, const uint64_t* sizes,
         __builtin_memcpy(rec, &off, 8);
         __builtin_memcpy(rec + 8, &m.key_size, 4);
         __builtin_memcpy(rec + 12, &m.full_size, 4);
-        src_map[p] = m.src;
+        src_map[p] =
+            (uint64_t)(R.data[m.src >> 48] + (m.src & 0xFFFFFFFFFFFFull));
     }
 }
 
-/* Balanced verbatim copy: 4-KiB destination window per 256-thread block,
- * one 16-B granule per thread. A window can intersect at most
- * 4096/32 + 2 entries (min entry size 32 B), so SPAN=136 offsets staged in
- * LDS always cover it; a 16-B granule straddles at most one entry boundary.
- */
+/* Balanced verbatim copy: a 16-KiB destination window per 256-thread
+ * block, 4 x 16-B granules per thread (independent loads/stores for ILP;
+ * granule passes are thread-contiguous so stores coalesce). Entries are
+ * >= 32 B (validated), so a window intersects < 16384/32 + 2 entries and
+ * SPAN=516 staged offsets always cover it; a 16-B granule straddles at
+ * most one entry boundary. */
 #define COPY_BLOCK 256
-#define COPY_WINDOW 4096
-#define COPY_SPAN 136
+#define COPY_WINDOW 16384
+#define COPY_SPAN 516
+
+#define COPY_GRANULES (COPY_WINDOW / 16)
+
+/* Per destination window, the largest survivor p with offset(p) <= window
+ * start — one thread per window (parallel), consumed by k_copy. */
+__global__ void k_winmap(const uint8_t* out_index, uint64_t n_surv,
+                         uint64_t total_bytes, uint32_t* win_p0) {
+    uint64_t n_windows = (total_bytes + COPY_WINDOW - 1) / COPY_WINDOW;
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t w = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         w < n_windows; w += stride) {
+        uint64_t wstart = w * COPY_WINDOW;
+        uint64_t lo = 0, hi = n_surv; /* first offset > wstart, minus 1 */
+        while (lo < hi) {
+            uint64_t mid = (lo + hi) >> 1;
+            if (ld_u64(out_index + mid * 16) <= wstart)
+                lo = mid + 1;
+            else
+                hi = mid;
+        }
+        win_p0[w] = (uint32_t)(lo - 1); /* offset(0)=0 -> lo >= 1 */
+    }
+}
 
 __global__ __launch_bounds__(COPY_BLOCK) void k_copy(
-    RunsDesc R, const uint8_t* out_index, const uint64_t* src_map,
-    uint64_t n_surv, uint64_t total_bytes, uint8_t* out_data) {
+    const uint8_t* out_index, const uint64_t* src_map,
+    const uint32_t* win_p0, uint64_t n_surv, uint64_t total_bytes,
+    uint8_t* out_data) {
     __shared__ uint64_t s_off[COPY_SPAN + 1];
     __shared__ uint64_t s_src[COPY_SPAN];
-    __shared__ uint64_t s_p0;
+    __shared__ uint16_t s_gid[COPY_GRANULES]; /* granule -> entry slot */
 
     uint64_t n_windows = (total_bytes + COPY_WINDOW - 1) / COPY_WINDOW;
     for (uint64_t w = blockIdx.x; w < n_windows; w += gridDim.x) {
         uint64_t wstart = w * COPY_WINDOW;
-        /* largest p with offset(p) <= wstart */
-        if (threadIdx.x == 0) {
-            uint64_t lo = 0, hi = n_surv; /* find first > wstart, minus 1 */
-            while (lo < hi) {
-                uint64_t mid = (lo + hi) >> 1;
-                uint64_t o = ld_u64(out_index + mid * 16);
-                if (o <= wstart)
-                    lo = mid + 1;
-                else
-                    hi = mid;
-            }
-            s_p0 = lo - 1; /* offset(0)=0 <= wstart always, so lo >= 1 */
-        }
-        __syncthreads();
-        uint64_t p0 = s_p0;
+        uint64_t p0 = win_p0[w];
         uint32_t cnt = (uint32_t)((n_surv - p0) < COPY_SPAN ? (n_surv - p0)
                                                             : COPY_SPAN);
-        if (threadIdx.x <= cnt && threadIdx.x < COPY_SPAN + 1) {
-            uint64_t p = p0 + threadIdx.x;
-            if (threadIdx.x == cnt)
+        for (uint32_t u = threadIdx.x; u <= cnt; u += COPY_BLOCK) {
+            uint64_t p = p0 + u;
+            if (u == cnt)
                 s_off[cnt] = (p < n_surv) ? ld_u64(out_index + p * 16)
                                           : total_bytes;
             else {
-                s_off[threadIdx.x] = ld_u64(out_index + p * 16);
-                s_src[threadIdx.x] = src_map[p];
+                s_off[u] = ld_u64(out_index + p * 16);
+                s_src[u] = src_map[p];
             }
         }
         __syncthreads();
+        /* fill the granule -> entry map: entry slot u owns granules whose
+         * START byte lies in [s_off[u], s_off[u+1]) */
+        for (uint32_t u = threadIdx.x; u < cnt; u += COPY_BLOCK) {
+            uint64_t b0 = s_off[u], b1 = s_off[u + 1];
+            uint64_t g0 = (b0 <= wstart) ? 0 : ((b0 - wstart + 15) >> 4);
+            uint64_t g1 = (b1 - wstart + 15) >> 4; /* exclusive */
+            if (g1 > COPY_GRANULES) g1 = COPY_GRANULES;
+            for (uint64_t g = g0; g < g1; g++) s_gid[g] = (uint16_t)u;
+        }
+        __syncthreads();
 
-        uint64_t gpos = wstart + (uint64_t)threadIdx.x * 16;
-        if (gpos < total_bytes) {
-            /* find j in [0, cnt): largest with s_off[j] <= gpos */
-            uint32_t lo = 0, hi = cnt;
-            while (lo < hi) {
-                uint32_t mid = (lo + hi) >> 1;
-                if (s_off[mid] <= gpos)
-                    lo = mid + 1;
-                else
-                    hi = mid;
-            }
-            uint32_t j = lo - 1;
+        #pragma unroll
+        for (int q = 0; q < 4; q++) {
+            uint32_t gl = q * COPY_BLOCK + threadIdx.x; /* window-local */
+            uint64_t gpos = wstart + (uint64_t)gl * 16;
+            if (gpos >= total_bytes) break;
+            uint32_t j = s_gid[gl];
             uint64_t e_end = s_off[j + 1];
-            uint64_t src_pack = s_src[j];
-            const uint8_t* src = R.data[src_pack >> 48] +
-                                 (src_pack & 0xFFFFFFFFFFFFull) +
+            const uint8_t* src = (const uint8_t*)s_src[j] +
                                  (gpos - s_off[j]);
             uint32_t nbytes =
                 (uint32_t)((total_bytes - gpos) < 16 ? (total_bytes - gpos)
@@ -385,9 +484,7 @@ __global__ __launch_bounds__(COPY_BLOCK) void k_copy(
                 /* one entry boundary inside the granule */
                 uint32_t c1 = (uint32_t)(e_end - gpos);
                 for (uint32_t b = 0; b < c1; b++) dst[b] = src[b];
-                uint64_t src2_pack = s_src[j + 1];
-                const uint8_t* src2 = R.data[src2_pack >> 48] +
-                                      (src2_pack & 0xFFFFFFFFFFFFull);
+                const uint8_t* src2 = (const uint8_t*)s_src[j + 1];
                 for (uint32_t b = c1; b < nbytes; b++)
                     dst[b] = src2[b - c1];
             }
@@ -417,6 +514,9 @@ struct dbeel_gpu_job {
     void* d_scantmp = nullptr;
     size_t scantmp_bytes = 0;
     uint32_t* d_err = nullptr;
+    uint64_t* d_pfx = nullptr;   /* dense big-endian key prefixes       */
+    uint32_t* d_piv = nullptr;   /* pivot crossranks                    */
+    uint32_t* d_winp0 = nullptr; /* copy window -> first survivor       */
     uint64_t total_entries = 0;
     uint64_t total_data_bytes = 0;
     uint64_t input_bytes = 0;
@@ -449,6 +549,10 @@ static int validate_runs(const dbeel_run_view* runs, size_t n_runs) {
         }
         if (runs[r].data_len >= (1ull << 48)) {
             set_err("run %zu: data_len too large", r);
+            return DBEEL_ERR_ITEM_TOO_LARGE;
+        }
+        if (runs[r].index_len / 16 >= (1ull << 32)) {
+            set_err("run %zu: more than 2^32 entries unsupported", r);
             return DBEEL_ERR_ITEM_TOO_LARGE;
         }
     }
@@ -519,6 +623,14 @@ extern "C" int dbeel_gpu_job_create(const dbeel_run_view* runs, size_t n_runs,
     JOB_CHECK(hipMalloc(&job->d_srcmap, n * sizeof(uint64_t)));
     JOB_CHECK(hipMalloc(&job->d_outdata, total_data ? total_data : 16));
     JOB_CHECK(hipMalloc(&job->d_err, 2 * sizeof(uint32_t)));
+    JOB_CHECK(hipMalloc(&job->d_pfx, n * sizeof(uint64_t)));
+    uint64_t n_piv_total = 0;
+    for (size_t r = 0; r < n_runs; r++)
+        n_piv_total += (runs[r].index_len / 16 + PIV_STRIDE - 1) >> PIV_SHIFT;
+    JOB_CHECK(hipMalloc(&job->d_piv,
+                        (n_piv_total ? n_piv_total : 1) * n_runs * 4));
+    JOB_CHECK(hipMalloc(&job->d_winp0,
+                        (total_data / COPY_WINDOW + 2) * sizeof(uint32_t)));
 
     size_t t1 = 0, t2 = 0;
     rocprim::exclusive_scan(nullptr, t1, job->d_sizes, job->d_dstoff,
@@ -556,6 +668,12 @@ extern "C" int dbeel_gpu_job_create(const dbeel_run_view* runs, size_t n_runs,
         D.entry_base[r] = base;
         base += D.count[r];
     }
+    uint32_t pbase = 0;
+    for (size_t r = 0; r < n_runs; r++) {
+        D.piv_base[r] = pbase;
+        pbase += (uint32_t)((D.count[r] + PIV_STRIDE - 1) >> PIV_SHIFT);
+    }
+    D.n_pivots = pbase;
     JOB_CHECK(hipEventRecord(job->ev[1], job->stream));
     JOB_CHECK(hipStreamSynchronize(job->stream));
     float ms = 0;
@@ -581,6 +699,9 @@ extern "C" void dbeel_gpu_job_destroy(dbeel_gpu_job* job) {
     hipFree(job->d_outdata);
     hipFree(job->d_scantmp);
     hipFree(job->d_err);
+    hipFree(job->d_pfx);
+    hipFree(job->d_piv);
+    hipFree(job->d_winp0);
     for (int i = 0; i < 8; i++)
         if (job->ev[i]) hipEventDestroy(job->ev[i]);
     if (job->stream) hipStreamDestroy(job->stream);
@@ -614,11 +735,20 @@ extern "C" int dbeel_gpu_job_run(dbeel_gpu_job* job, int keep_tombstones,
     HIP_CHECK(hipEventRecord(job->ev[0], s));
     if (n) {
         uint32_t grid = pick_grid(n, 256);
-        hipLaunchKernelGGL(k_validate, dim3(grid), dim3(256), 0, s, job->desc,
-                           job->d_err);
+        hipLaunchKernelGGL(k_prepare, dim3(grid), dim3(256), 0, s, job->desc,
+                           job->d_pfx, job->d_err);
+        if (job->desc.n_pivots)
+            hipLaunchKernelGGL(k_pivots,
+                               dim3(pick_grid(job->desc.n_pivots, 256)),
+                               dim3(256), 0, s, job->desc, job->d_pfx,
+                               job->d_piv);
+    }
+    HIP_CHECK(hipEventRecord(job->ev[6], s));
+    if (n) {
+        uint32_t grid = pick_grid(n, 256);
         hipLaunchKernelGGL(k_rank, dim3(grid), dim3(256), 0, s, job->desc,
-                           job->d_meta, job->d_sizes, job->d_flags,
-                           keep_tombstones);
+                           job->d_pfx, job->d_piv, job->d_meta, job->d_sizes,
+                           job->d_flags, keep_tombstones);
     }
     HIP_CHECK(hipEventRecord(job->ev[1], s));
     if (n) {
@@ -633,9 +763,9 @@ extern "C" int dbeel_gpu_job_run(dbeel_gpu_job* job, int keep_tombstones,
     HIP_CHECK(hipEventRecord(job->ev[2], s));
     if (n) {
         uint32_t grid = pick_grid(n, 256);
-        hipLaunchKernelGGL(k_emit, dim3(grid), dim3(256), 0, s, job->d_meta,
-                           job->d_sizes, job->d_dstoff, job->d_pos, n,
-                           job->d_outindex, job->d_srcmap);
+        hipLaunchKernelGGL(k_emit, dim3(grid), dim3(256), 0, s, job->desc,
+                           job->d_meta, job->d_sizes, job->d_dstoff,
+                           job->d_pos, n, job->d_outindex, job->d_srcmap);
     }
     HIP_CHECK(hipEventRecord(job->ev[3], s));
 
@@ -668,9 +798,12 @@ extern "C" int dbeel_gpu_job_run(dbeel_gpu_job* job, int keep_tombstones,
     if (total_out) {
         uint64_t windows = (total_out + COPY_WINDOW - 1) / COPY_WINDOW;
         uint32_t grid = windows > 4096 ? 4096 : (uint32_t)windows;
+        hipLaunchKernelGGL(k_winmap, dim3(pick_grid(windows, 256)), dim3(256),
+                           0, s, job->d_outindex, n_surv, total_out,
+                           job->d_winp0);
         hipLaunchKernelGGL(k_copy, dim3(grid), dim3(COPY_BLOCK), 0, s,
-                           job->desc, job->d_outindex, job->d_srcmap, n_surv,
-                           total_out, job->d_outdata);
+                           job->d_outindex, job->d_srcmap, job->d_winp0,
+                           n_surv, total_out, job->d_outdata);
     }
     HIP_CHECK(hipEventRecord(job->ev[5], s));
     HIP_CHECK(hipStreamSynchronize(s));
@@ -682,17 +815,20 @@ extern "C" int dbeel_gpu_job_run(dbeel_gpu_job* job, int keep_tombstones,
     if (out_data_len) *out_data_len = total_out;
     if (out_entries) *out_entries = n_surv;
     if (t) {
-        float rank_ms = 0, scan_ms = 0, emit_ms = 0, copy_ms = 0;
-        hipEventElapsedTime(&rank_ms, job->ev[0], job->ev[1]);
-        hipEventElapsedTime(&scan_ms, job->ev[1], job->ev[2]);
-        hipEventElapsedTime(&emit_ms, job->ev[2], job->ev[3]);
-        hipEventElapsedTime(&copy_ms, job->ev[4], job->ev[5]);
+        float prep_ms = 0, rank_ms = 0, scan_ms = 0, emit_ms = 0,
+              copy_ms = 0;
+        (void)hipEventElapsedTime(&prep_ms, job->ev[0], job->ev[6]);
+        (void)hipEventElapsedTime(&rank_ms, job->ev[6], job->ev[1]);
+        (void)hipEventElapsedTime(&scan_ms, job->ev[1], job->ev[2]);
+        (void)hipEventElapsedTime(&emit_ms, job->ev[2], job->ev[3]);
+        (void)hipEventElapsedTime(&copy_ms, job->ev[4], job->ev[5]);
         t->h2d_ms = job->h2d_ms;
+        t->prep_ms = prep_ms;
         t->rank_ms = rank_ms;
         t->scan_ms = scan_ms;
         t->emit_ms = emit_ms;
         t->copy_ms = copy_ms;
-        t->kernel_ms = rank_ms + scan_ms + emit_ms + copy_ms;
+        t->kernel_ms = prep_ms + rank_ms + scan_ms + emit_ms + copy_ms;
         t->d2h_ms = 0.0;
     }
     return DBEEL_OK;

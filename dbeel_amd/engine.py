@@ -53,6 +53,7 @@ class CompactResult(ctypes.Structure):
 class CompactTimings(ctypes.Structure):
     _fields_ = [
         ("h2d_ms", ctypes.c_double),
+        ("prep_ms", ctypes.c_double),
         ("rank_ms", ctypes.c_double),
         ("scan_ms", ctypes.c_double),
         ("emit_ms", ctypes.c_double),

@@ -70,11 +70,13 @@ typedef struct {
  * (rank -> scan -> emit -> copy), excluding host<->device copies. */
 typedef struct {
     double h2d_ms;
+    double prep_ms;    /* k_prepare: validation + key-prefix extract,
+                          k_pivots: crossrank pivots                 */
     double rank_ms;    /* k_rank: global merge-rank + winner flags  */
     double scan_ms;    /* size/position prefix sums                 */
     double emit_ms;    /* output index build + survivor source map  */
     double copy_ms;    /* verbatim entry copy-out                   */
-    double kernel_ms;  /* rank+scan+emit+copy                       */
+    double kernel_ms;  /* prep+rank+scan+emit+copy                  */
     double d2h_ms;
 } dbeel_compact_timings;
 
