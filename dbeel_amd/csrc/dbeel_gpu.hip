@@ -313,57 +313,153 @@ __global__ void k_pivots(RunsDesc R, const uint64_t* pfx, uint32_t* piv) {
     }
 }
 
+/* Lower bound by linear advance from a known smaller bound (the previous
+ * consecutive entry's crossrank) with a binary-search fallback for skewed
+ * key distributions. Semantics identical to lower_rank. */
+__device__ __forceinline__ uint64_t advance_rank(const RunsDesc& R,
+                                                 const uint64_t* pfx, int r2,
+                                                 const EView& e,
+                                                 uint64_t epfx, int re,
+                                                 uint64_t lo, uint64_t hi,
+                                                 bool& equal_key_at) {
+    const uint64_t* p2 = pfx + R.entry_base[r2];
+    uint64_t cur = lo;
+    int steps = 0;
+    while (cur < hi) {
+        uint64_t pm = p2[cur];
+        int c;
+        if (pm != epfx) {
+            c = (pm < epfx) ? -1 : 1;
+        } else {
+            EView m;
+            if (!load_entry(R, r2, cur, m)) break;
+            c = cmp_full(m, r2, e, re);
+        }
+        if (c >= 0) break;
+        cur++;
+        if (++steps > 24)
+            return lower_rank(R, pfx, r2, e, epfx, re, cur, hi,
+                              equal_key_at);
+    }
+    equal_key_at = false;
+    if (cur < R.count[r2] && p2[cur] == epfx) {
+        EView m;
+        if (load_entry(R, r2, cur, m))
+            equal_key_at = (m.klen == e.klen) &&
+                           (cmp_keys(m.key, m.klen, e.key, e.klen) == 0);
+    }
+    return cur;
+}
+
+/* Entries ranked RANK_BATCH consecutive per thread: entry b=0 binary
+ * searches within its pivot window; b>0 advance linearly from b-1's
+ * crossrank (crossranks are monotone within a run). */
+#define RANK_BATCH 4
+
 __global__ void k_rank(RunsDesc R, const uint64_t* pfx, const uint32_t* piv,
                        EntryMeta* meta, uint64_t* sizes, uint32_t* flags,
                        int keep_tombstones) {
-    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
-    for (uint64_t g = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
-         g < R.total; g += stride) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x * RANK_BATCH;
+    for (uint64_t g0 = ((uint64_t)blockIdx.x * blockDim.x + threadIdx.x) *
+                       RANK_BATCH;
+         g0 < R.total; g0 += stride) {
         int r = 0;
-        while (r + 1 < R.n_runs && g >= R.entry_base[r + 1]) r++;
-        uint64_t i = g - R.entry_base[r];
-        EView e;
-        if (!load_entry(R, r, i, e)) {
-            /* k_prepare has flagged this input; keep memory safe and park
-             * the entry at its local slot (results will be discarded) */
-            meta[g] = EntryMeta{0, 8, 32};
-            sizes[g] = 0;
-            flags[g] = 0;
-            continue;
-        }
-        uint64_t epfx = pfx[g];
+        while (r + 1 < R.n_runs && g0 >= R.entry_base[r + 1]) r++;
+        uint64_t run_end = R.entry_base[r] + R.count[r];
+        /* batch stays inside one run; a batch crossing the boundary is
+         * truncated and the tail re-dispatched from the next run */
+        uint64_t gmax = g0 + RANK_BATCH;
+        if (gmax > run_end) gmax = run_end;
+        if (gmax > R.total) gmax = R.total;
 
-        /* pivot window of this entry: search bounds per other run */
-        uint32_t j = (uint32_t)(i >> PIV_SHIFT);
-        uint32_t n_piv_r =
-            (uint32_t)((R.count[r] + PIV_STRIDE - 1) >> PIV_SHIFT);
-        const uint32_t* lo_cr = piv + ((uint64_t)R.piv_base[r] + j) * R.n_runs;
-        const uint32_t* hi_cr =
-            (j + 1 < n_piv_r)
-                ? piv + ((uint64_t)R.piv_base[r] + j + 1) * R.n_runs
-                : nullptr;
+        for (uint64_t gseg = g0; gseg < g0 + RANK_BATCH;) {
+            int nb = (int)(gmax - gseg);
+            if (nb <= 0) {
+                /* tail crossed into the next run */
+                if (gseg >= R.total) break;
+                while (r + 1 < R.n_runs && gseg >= R.entry_base[r + 1]) r++;
+                run_end = R.entry_base[r] + R.count[r];
+                gmax = g0 + RANK_BATCH;
+                if (gmax > run_end) gmax = run_end;
+                if (gmax > R.total) gmax = R.total;
+                continue;
+            }
 
-        uint64_t rank = i;
-        bool winner = true;
-        for (int r2 = 0; r2 < R.n_runs; r2++) {
-            if (r2 == r) continue;
-            uint64_t lo = lo_cr[r2];
-            uint64_t hi = hi_cr ? (uint64_t)hi_cr[r2] : R.count[r2];
-            bool eq;
-            rank += lower_rank(R, pfx, r2, e, epfx, r, lo, hi, eq);
-            /* an equal key at the insertion point is same-key and later in
-             * the order -> it supersedes e (newest-wins dedup) */
-            winner &= !eq;
+            uint64_t i0 = gseg - R.entry_base[r];
+            EView e[RANK_BATCH];
+            uint64_t epfx[RANK_BATCH];
+            uint64_t rank[RANK_BATCH];
+            bool winner[RANK_BATCH];
+            bool ok = true;
+            #pragma unroll
+            for (int b = 0; b < RANK_BATCH; b++) {
+                if (b >= nb) break;
+                if (!load_entry(R, r, i0 + b, e[b])) ok = false;
+                epfx[b] = pfx[gseg + b];
+                rank[b] = i0 + b;
+                winner[b] = true;
+            }
+            if (!ok) {
+                /* k_prepare flagged this input; park entries at their
+                 * local slots, results are discarded on the host */
+                #pragma unroll
+                for (int b = 0; b < RANK_BATCH; b++) {
+                    if (b >= nb) break;
+                    meta[gseg + b] = EntryMeta{0, 8, 32};
+                    sizes[gseg + b] = 0;
+                    flags[gseg + b] = 0;
+                }
+                gseg += nb;
+                continue;
+            }
+
+            /* pivot windows of first and last batch entry */
+            uint32_t j0 = (uint32_t)(i0 >> PIV_SHIFT);
+            uint32_t j1 = (uint32_t)((i0 + nb - 1) >> PIV_SHIFT);
+            uint32_t n_piv_r =
+                (uint32_t)((R.count[r] + PIV_STRIDE - 1) >> PIV_SHIFT);
+            const uint32_t* lo_cr =
+                piv + ((uint64_t)R.piv_base[r] + j0) * R.n_runs;
+            const uint32_t* hi_cr =
+                (j1 + 1 < n_piv_r)
+                    ? piv + ((uint64_t)R.piv_base[r] + j1 + 1) * R.n_runs
+                    : nullptr;
+
+            for (int r2 = 0; r2 < R.n_runs; r2++) {
+                if (r2 == r) continue;
+                uint64_t lo = lo_cr[r2];
+                uint64_t hi = hi_cr ? (uint64_t)hi_cr[r2] : R.count[r2];
+                bool eq;
+                uint64_t cr =
+                    lower_rank(R, pfx, r2, e[0], epfx[0], r, lo, hi, eq);
+                rank[0] += cr;
+                winner[0] &= !eq;
+                #pragma unroll
+                for (int b = 1; b < RANK_BATCH; b++) {
+                    if (b >= nb) break;
+                    cr = advance_rank(R, pfx, r2, e[b], epfx[b], r, cr, hi,
+                                      eq);
+                    rank[b] += cr;
+                    winner[b] &= !eq;
+                }
+            }
+
+            #pragma unroll
+            for (int b = 0; b < RANK_BATCH; b++) {
+                if (b >= nb) break;
+                uint64_t dlen = (uint64_t)e[b].full_size - 32 - e[b].klen;
+                bool keep =
+                    winner[b] && (keep_tombstones || dlen != 0);
+                EntryMeta m;
+                m.src = ((uint64_t)r << 48) | e[b].off;
+                m.key_size = e[b].key_size;
+                m.full_size = e[b].full_size;
+                meta[rank[b]] = m;
+                sizes[rank[b]] = keep ? e[b].full_size : 0;
+                flags[rank[b]] = keep ? 1u : 0u;
+            }
+            gseg += nb;
         }
-        uint64_t dlen = (uint64_t)e.full_size - 32 - e.klen;
-        bool keep = winner && (keep_tombstones || dlen != 0);
-        EntryMeta m;
-        m.src = ((uint64_t)r << 48) | e.off;
-        m.key_size = e.key_size;
-        m.full_size = e.full_size;
-        meta[rank] = m;
-        sizes[rank] = keep ? e.full_size : 0;
-        flags[rank] = keep ? 1u : 0u;
     }
 }
 
@@ -435,8 +531,14 @@ __global__ __launch_bounds__(COPY_BLOCK) void k_copy(
     for (uint64_t w = blockIdx.x; w < n_windows; w += gridDim.x) {
         uint64_t wstart = w * COPY_WINDOW;
         uint64_t p0 = win_p0[w];
-        uint32_t cnt = (uint32_t)((n_surv - p0) < COPY_SPAN ? (n_surv - p0)
-                                                            : COPY_SPAN);
+        /* entries intersecting this window are p0 .. win_p0[w+1]; staging
+         * only those (not the worst-case SPAN) saves ~30x index re-reads
+         * on KiB-sized entries */
+        uint64_t p_end = (w + 1 < n_windows) ? (uint64_t)win_p0[w + 1] + 1
+                                             : n_surv;
+        uint32_t cnt = (uint32_t)(p_end - p0);
+        if (cnt > COPY_SPAN) cnt = COPY_SPAN;
+        if (cnt > n_surv - p0) cnt = (uint32_t)(n_surv - p0);
         for (uint32_t u = threadIdx.x; u <= cnt; u += COPY_BLOCK) {
             uint64_t p = p0 + u;
             if (u == cnt)
