@@ -74,17 +74,9 @@ struct RunsDesc {
     uint64_t data_len[MAX_RUNS];
     uint64_t count[MAX_RUNS];       /* entries per run                  */
     uint64_t entry_base[MAX_RUNS];  /* exclusive prefix sum of count    */
-    uint32_t piv_base[MAX_RUNS];    /* exclusive prefix sum of n pivots */
     int n_runs;
     uint64_t total;                 /* sum of count                     */
-    uint32_t n_pivots;              /* total pivots over all runs       */
 };
-
-/* Pivot stride: crossranks of every PIV_STRIDE-th entry of each run into
- * every other run are precomputed (k_pivots), bounding each entry's
- * binary search to a ~PIV_STRIDE-wide, L2-resident window. */
-#define PIV_SHIFT 10
-#define PIV_STRIDE (1u << PIV_SHIFT)
 
 /* 16 B/entry scratch record, scattered by global rank in k_rank. */
 struct EntryMeta {
@@ -204,11 +196,77 @@ __device__ __forceinline__ uint64_t key_prefix(const uint8_t* key,
     return __builtin_bswap64(v);
 }
 
+/* Dense per-entry search record (one cache line for two): comparisons on
+ * prefix-tied entries read THIS instead of the scattered multi-GiB data
+ * blob. Keys longer than AUX_KEY_BYTES fall back to the blob only when two
+ * such keys agree on their first AUX_KEY_BYTES bytes (true duplicates of
+ * long keys). */
+#define AUX_KEY_BYTES 40
+struct Aux {
+    uint32_t klen;
+    uint32_t rsv;
+    uint64_t ts_lo; /* timestamp i128 LE halves */
+    int64_t ts_hi;
+    uint8_t key[AUX_KEY_BYTES]; /* zero-padded */
+};
+static_assert(sizeof(Aux) == 64, "aux record must be 64B");
+
+/* Key-only compare of entries A=(rA,iA), B=(rB,iB) via aux records. */
+__device__ __forceinline__ int cmp_keys_aux(const RunsDesc& R,
+                                            const Aux* aux, int rA,
+                                            uint64_t iA, int rB,
+                                            uint64_t iB) {
+    const Aux* a = aux + R.entry_base[rA] + iA;
+    const Aux* b = aux + R.entry_base[rB] + iB;
+    uint32_t la = a->klen, lb = b->klen;
+    uint32_t n = la < lb ? la : lb;
+    if (n > AUX_KEY_BYTES) n = AUX_KEY_BYTES;
+    #pragma unroll
+    for (uint32_t i = 0; i < AUX_KEY_BYTES; i += 8) {
+        if (i >= n) break;
+        uint64_t va = ld_u64(a->key + i), vb = ld_u64(b->key + i);
+        if (i + 8 > n) { /* mask the tail (pad bytes are zero anyway) */
+            uint64_t mask = (~0ull) >> (8 * (i + 8 - n));
+            va &= mask;
+            vb &= mask;
+        }
+        if (va != vb) {
+            va = __builtin_bswap64(va);
+            vb = __builtin_bswap64(vb);
+            return va < vb ? -1 : 1;
+        }
+    }
+    if (la > AUX_KEY_BYTES && lb > AUX_KEY_BYTES) {
+        /* rare: long keys tied on the first AUX_KEY_BYTES bytes */
+        EView ea, eb;
+        if (!load_entry(R, rA, iA, ea) || !load_entry(R, rB, iB, eb))
+            return -1; /* corrupt input flagged elsewhere; value discarded */
+        return cmp_keys(ea.key, ea.klen, eb.key, eb.klen);
+    }
+    return la < lb ? -1 : (la > lb ? 1 : 0);
+}
+
+/* Full-order compare (key, timestamp, run index — lsm_tree.rs:52-71) via
+ * aux records. */
+__device__ __forceinline__ int cmp_aux_full(const RunsDesc& R,
+                                            const Aux* aux, int rA,
+                                            uint64_t iA, int rB,
+                                            uint64_t iB) {
+    int c = cmp_keys_aux(R, aux, rA, iA, rB, iB);
+    if (c) return c;
+    const Aux* a = aux + R.entry_base[rA] + iA;
+    const Aux* b = aux + R.entry_base[rB] + iB;
+    if (a->ts_hi != b->ts_hi) return a->ts_hi < b->ts_hi ? -1 : 1;
+    if (a->ts_lo != b->ts_lo) return a->ts_lo < b->ts_lo ? -1 : 1;
+    return rA < rB ? -1 : (rA > rB ? 1 : 0);
+}
+
 /* Validates every entry (bounds + bincode field cross-check,
- * read_next_entry lsm_tree.rs:1158-70), checks each run is strictly sorted
- * by key (flush invariant), and extracts the dense key-prefix array the
- * rank searches run on. */
-__global__ void k_prepare(RunsDesc R, uint64_t* pfx, uint32_t* err) {
+ * read_next_entry lsm_tree.rs:1158-70) and extracts the dense key-prefix
+ * and aux arrays the rank searches run on. Run sortedness is checked in
+ * k_rank (it already holds consecutive entries). */
+__global__ void k_prepare(RunsDesc R, uint64_t* pfx, Aux* aux,
+                          uint32_t* err) {
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
     for (uint64_t g = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
          g < R.total; g += stride) {
@@ -220,246 +278,184 @@ __global__ void k_prepare(RunsDesc R, uint64_t* pfx, uint32_t* err) {
         if (!load_entry(R, r, i, e)) {
             atomicOr(err, DERR_CORRUPT);
             pfx[g] = 0;
+            Aux z = {};
+            aux[g] = z;
             continue;
         }
         pfx[g] = key_prefix(e.key, e.klen);
+        Aux a;
+        a.klen = (uint32_t)e.klen;
+        a.rsv = 0;
+        load_ts(e, a.ts_lo, a.ts_hi);
+        uint32_t nk = e.klen < AUX_KEY_BYTES ? (uint32_t)e.klen
+                                             : AUX_KEY_BYTES;
+        #pragma unroll
+        for (uint32_t b = 0; b < AUX_KEY_BYTES; b++)
+            a.key[b] = (b < nk) ? e.key[b] : 0;
+        aux[g] = a;
         /* bincode field cross-check */
         if (ld_u64(e.raw) != e.klen ||
-            ld_u64(e.raw + 8 + e.klen) != (uint64_t)e.full_size - 32 - e.klen) {
+            ld_u64(e.raw + 8 + e.klen) != (uint64_t)e.full_size - 32 - e.klen)
             atomicOr(err, DERR_CORRUPT);
-            continue;
-        }
-        /* run must be strictly sorted by key (flush invariant) */
-        if (i + 1 < R.count[r]) {
-            EView nx;
-            if (!load_entry(R, r, i + 1, nx)) {
-                atomicOr(err, DERR_CORRUPT);
-                continue;
-            }
-            if (cmp_keys(e.key, e.klen, nx.key, nx.klen) >= 0)
-                atomicOr(err, DERR_UNSORTED);
-        }
     }
 }
 
-/* Number of entries in run r2 (within [lo, hi)) strictly below e (full
- * order). Runs are sorted by key with unique keys, hence sorted in full
- * order too. Search steps compare dense 8-byte prefixes first; the full
- * comparator (scattered entry read) runs only on prefix ties. */
-__device__ __forceinline__ uint64_t lower_rank(const RunsDesc& R,
-                                               const uint64_t* pfx, int r2,
-                                               const EView& e, uint64_t epfx,
-                                               int re, uint64_t lo,
-                                               uint64_t hi,
-                                               bool& equal_key_at) {
-    const uint64_t* p2 = pfx + R.entry_base[r2];
-    while (lo < hi) {
-        uint64_t mid = (lo + hi) >> 1;
-        uint64_t pm = p2[mid];
-        if (pm != epfx) {
-            if (pm < epfx)
+/* ------------------------------------------------------------------ */
+/* Crossranks by merge-path co-ranking                                 */
+/*                                                                     */
+/* For every run pair (a, b), the merged sequence (full order: key,    */
+/* timestamp, run index — a strict total order, so no merge ties) is   */
+/* partitioned into CORANK_CHUNK-wide chunks by diagonal binary        */
+/* search; each chunk is walked linearly, recording for every consumed */
+/* entry its crossrank into the opposite run (= the opposite cursor)   */
+/* and whether the opposite run's next entry carries the same key      */
+/* (then it is later in the order and supersedes it — the newest-wins  */
+/* dedup of lsm_tree.rs:1041-1046). Work is (k-1)*N sequential,        */
+/* cache-resident compares — not N*(k-1)*log2(N) random probes.        */
+/*                                                                     */
+/* cr layout: column-major, cr[s * total + g] = crossrank of entry g   */
+/* into the s-th OTHER run of g's run (s = r2 < r ? r2 : r2-1), with   */
+/* bit 31 = superseded flag. Every slot is written exactly once.       */
+/* ------------------------------------------------------------------ */
+
+#define CORANK_CHUNK 128
+#define CR_LOSER 0x80000000u
+#define CR_MASK 0x7FFFFFFFu
+
+struct PairDesc {
+    uint32_t a, b;
+    uint64_t chunk_base; /* exclusive prefix sum of per-pair chunks */
+};
+
+/* strict full-order compare of a[ia] vs b[ib] on dense arrays */
+__device__ __forceinline__ int cmp_pair(const RunsDesc& R,
+                                        const uint64_t* pfx, const Aux* aux,
+                                        int ra, uint64_t ia, int rb,
+                                        uint64_t ib) {
+    uint64_t pa = pfx[R.entry_base[ra] + ia];
+    uint64_t pb = pfx[R.entry_base[rb] + ib];
+    if (pa != pb) return pa < pb ? -1 : 1;
+    return cmp_aux_full(R, aux, ra, ia, rb, ib);
+}
+
+__device__ __forceinline__ bool keys_eq_pair(const RunsDesc& R,
+                                             const uint64_t* pfx,
+                                             const Aux* aux, int ra,
+                                             uint64_t ia, int rb,
+                                             uint64_t ib) {
+    if (pfx[R.entry_base[ra] + ia] != pfx[R.entry_base[rb] + ib])
+        return false;
+    return cmp_keys_aux(R, aux, ra, ia, rb, ib) == 0;
+}
+
+__global__ void k_corank(RunsDesc R, const uint64_t* pfx, const Aux* aux,
+                         const PairDesc* pairs, uint32_t n_pairs,
+                         uint64_t total_chunks, uint32_t* cr) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t t = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         t < total_chunks; t += stride) {
+        /* largest pair with chunk_base <= t */
+        uint32_t lo = 0, hi = n_pairs;
+        while (lo < hi) {
+            uint32_t mid = (lo + hi) >> 1;
+            if (pairs[mid].chunk_base <= t)
                 lo = mid + 1;
             else
                 hi = mid;
-            continue;
         }
-        EView m;
-        /* corrupt records are flagged by k_prepare; avoid the OOB read
-         * here (results are discarded once the host sees the flag) */
-        if (!load_entry(R, r2, mid, m)) {
-            hi = mid;
-            continue;
+        const PairDesc P = pairs[lo - 1];
+        const int a = (int)P.a, b = (int)P.b;
+        const uint64_t Na = R.count[a], Nb = R.count[b];
+        uint64_t diag = (t - P.chunk_base) * CORANK_CHUNK;
+        uint64_t end = diag + CORANK_CHUNK;
+        if (end > Na + Nb) end = Na + Nb;
+
+        /* diagonal search: ia = # of a-entries among the first `diag`
+         * merged entries */
+        uint64_t slo = diag > Nb ? diag - Nb : 0;
+        uint64_t shi = diag < Na ? diag : Na;
+        while (slo < shi) {
+            uint64_t mid = (slo + shi) >> 1;
+            if (cmp_pair(R, pfx, aux, a, mid, b, diag - mid - 1) < 0)
+                slo = mid + 1;
+            else
+                shi = mid;
         }
-        if (cmp_full(m, r2, e, re) < 0)
-            lo = mid + 1;
-        else
-            hi = mid;
+        uint64_t ia = slo, ib = diag - slo;
+
+        uint32_t sa = (uint32_t)(b < a ? b : b - 1); /* slot of b in a */
+        uint32_t sb = (uint32_t)(a < b ? a : a - 1); /* slot of a in b */
+        uint32_t* cra = cr + (uint64_t)sa * R.total + R.entry_base[a];
+        uint32_t* crb = cr + (uint64_t)sb * R.total + R.entry_base[b];
+
+        for (uint64_t pos = diag; pos < end; pos++) {
+            bool take_a;
+            if (ia >= Na)
+                take_a = false;
+            else if (ib >= Nb)
+                take_a = true;
+            else
+                take_a = cmp_pair(R, pfx, aux, a, ia, b, ib) < 0;
+            if (take_a) {
+                uint32_t v = (uint32_t)ib;
+                if (ib < Nb && keys_eq_pair(R, pfx, aux, a, ia, b, ib))
+                    v |= CR_LOSER;
+                cra[ia] = v;
+                ia++;
+            } else {
+                uint32_t v = (uint32_t)ia;
+                if (ia < Na && keys_eq_pair(R, pfx, aux, b, ib, a, ia))
+                    v |= CR_LOSER;
+                crb[ib] = v;
+                ib++;
+            }
+        }
     }
-    equal_key_at = false;
-    if (lo < R.count[r2] && p2[lo] == epfx) {
-        EView m;
-        if (load_entry(R, r2, lo, m))
-            equal_key_at = (m.klen == e.klen) &&
-                           (cmp_keys(m.key, m.klen, e.key, e.klen) == 0);
-    }
-    return lo;
 }
 
-/* Crossranks of every PIV_STRIDE-th entry of each run into every other
- * run: piv[(piv_base[r] + j) * n_runs + r2] = lower_rank of entry
- * (r, j*PIV_STRIDE) in run r2. Tiny kernel (N/1024 x n_runs searches). */
-__global__ void k_pivots(RunsDesc R, const uint64_t* pfx, uint32_t* piv) {
-    uint32_t stride = gridDim.x * blockDim.x;
-    for (uint32_t t = blockIdx.x * blockDim.x + threadIdx.x; t < R.n_pivots;
-         t += stride) {
+/* Reduce per-pair crossranks to the global rank, apply the winner /
+ * tombstone rules, and emit the rank-indexed scratch records. Also checks
+ * each run is strictly sorted by key (flush invariant,
+ * lsm_tree.rs:925-946) on the dense aux records. */
+__global__ void k_rankreduce(RunsDesc R, const Aux* aux, const uint32_t* cr,
+                             EntryMeta* meta, uint64_t* sizes,
+                             uint32_t* flags, int keep_tombstones,
+                             uint32_t* err) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t g = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         g < R.total; g += stride) {
         int r = 0;
-        while (r + 1 < R.n_runs && t >= R.piv_base[r + 1]) r++;
-        uint64_t i = (uint64_t)(t - R.piv_base[r]) << PIV_SHIFT;
+        while (r + 1 < R.n_runs && g >= R.entry_base[r + 1]) r++;
+        uint64_t i = g - R.entry_base[r];
         EView e;
         if (!load_entry(R, r, i, e)) {
-            for (int r2 = 0; r2 < R.n_runs; r2++)
-                piv[(uint64_t)t * R.n_runs + r2] = 0;
+            /* k_prepare has flagged this input; keep memory safe and park
+             * the entry at its local slot (results will be discarded) */
+            meta[g] = EntryMeta{0, 8, 32};
+            sizes[g] = 0;
+            flags[g] = 0;
             continue;
         }
-        uint64_t epfx = key_prefix(e.key, e.klen);
-        for (int r2 = 0; r2 < R.n_runs; r2++) {
-            uint64_t cr = 0;
-            if (r2 != r) {
-                bool eq;
-                cr = lower_rank(R, pfx, r2, e, epfx, r, 0, R.count[r2], eq);
-            }
-            piv[(uint64_t)t * R.n_runs + r2] = (uint32_t)cr;
+        if (i + 1 < R.count[r] &&
+            cmp_keys_aux(R, aux, r, i, r, i + 1) >= 0)
+            atomicOr(err, DERR_UNSORTED);
+
+        uint64_t rank = i;
+        bool winner = true;
+        for (int s = 0; s + 1 < R.n_runs; s++) {
+            uint32_t v = cr[(uint64_t)s * R.total + g];
+            rank += v & CR_MASK;
+            winner &= !(v & CR_LOSER);
         }
-    }
-}
-
-/* Lower bound by linear advance from a known smaller bound (the previous
- * consecutive entry's crossrank) with a binary-search fallback for skewed
- * key distributions. Semantics identical to lower_rank. */
-__device__ __forceinline__ uint64_t advance_rank(const RunsDesc& R,
-                                                 const uint64_t* pfx, int r2,
-                                                 const EView& e,
-                                                 uint64_t epfx, int re,
-                                                 uint64_t lo, uint64_t hi,
-                                                 bool& equal_key_at) {
-    const uint64_t* p2 = pfx + R.entry_base[r2];
-    uint64_t cur = lo;
-    int steps = 0;
-    while (cur < hi) {
-        uint64_t pm = p2[cur];
-        int c;
-        if (pm != epfx) {
-            c = (pm < epfx) ? -1 : 1;
-        } else {
-            EView m;
-            if (!load_entry(R, r2, cur, m)) break;
-            c = cmp_full(m, r2, e, re);
-        }
-        if (c >= 0) break;
-        cur++;
-        if (++steps > 24)
-            return lower_rank(R, pfx, r2, e, epfx, re, cur, hi,
-                              equal_key_at);
-    }
-    equal_key_at = false;
-    if (cur < R.count[r2] && p2[cur] == epfx) {
-        EView m;
-        if (load_entry(R, r2, cur, m))
-            equal_key_at = (m.klen == e.klen) &&
-                           (cmp_keys(m.key, m.klen, e.key, e.klen) == 0);
-    }
-    return cur;
-}
-
-/* Entries ranked RANK_BATCH consecutive per thread: entry b=0 binary
- * searches within its pivot window; b>0 advance linearly from b-1's
- * crossrank (crossranks are monotone within a run). */
-#define RANK_BATCH 4
-
-__global__ void k_rank(RunsDesc R, const uint64_t* pfx, const uint32_t* piv,
-                       EntryMeta* meta, uint64_t* sizes, uint32_t* flags,
-                       int keep_tombstones) {
-    uint64_t stride = (uint64_t)gridDim.x * blockDim.x * RANK_BATCH;
-    for (uint64_t g0 = ((uint64_t)blockIdx.x * blockDim.x + threadIdx.x) *
-                       RANK_BATCH;
-         g0 < R.total; g0 += stride) {
-        int r = 0;
-        while (r + 1 < R.n_runs && g0 >= R.entry_base[r + 1]) r++;
-        uint64_t run_end = R.entry_base[r] + R.count[r];
-        /* batch stays inside one run; a batch crossing the boundary is
-         * truncated and the tail re-dispatched from the next run */
-        uint64_t gmax = g0 + RANK_BATCH;
-        if (gmax > run_end) gmax = run_end;
-        if (gmax > R.total) gmax = R.total;
-
-        for (uint64_t gseg = g0; gseg < g0 + RANK_BATCH;) {
-            int nb = (int)(gmax - gseg);
-            if (nb <= 0) {
-                /* tail crossed into the next run */
-                if (gseg >= R.total) break;
-                while (r + 1 < R.n_runs && gseg >= R.entry_base[r + 1]) r++;
-                run_end = R.entry_base[r] + R.count[r];
-                gmax = g0 + RANK_BATCH;
-                if (gmax > run_end) gmax = run_end;
-                if (gmax > R.total) gmax = R.total;
-                continue;
-            }
-
-            uint64_t i0 = gseg - R.entry_base[r];
-            EView e[RANK_BATCH];
-            uint64_t epfx[RANK_BATCH];
-            uint64_t rank[RANK_BATCH];
-            bool winner[RANK_BATCH];
-            bool ok = true;
-            #pragma unroll
-            for (int b = 0; b < RANK_BATCH; b++) {
-                if (b >= nb) break;
-                if (!load_entry(R, r, i0 + b, e[b])) ok = false;
-                epfx[b] = pfx[gseg + b];
-                rank[b] = i0 + b;
-                winner[b] = true;
-            }
-            if (!ok) {
-                /* k_prepare flagged this input; park entries at their
-                 * local slots, results are discarded on the host */
-                #pragma unroll
-                for (int b = 0; b < RANK_BATCH; b++) {
-                    if (b >= nb) break;
-                    meta[gseg + b] = EntryMeta{0, 8, 32};
-                    sizes[gseg + b] = 0;
-                    flags[gseg + b] = 0;
-                }
-                gseg += nb;
-                continue;
-            }
-
-            /* pivot windows of first and last batch entry */
-            uint32_t j0 = (uint32_t)(i0 >> PIV_SHIFT);
-            uint32_t j1 = (uint32_t)((i0 + nb - 1) >> PIV_SHIFT);
-            uint32_t n_piv_r =
-                (uint32_t)((R.count[r] + PIV_STRIDE - 1) >> PIV_SHIFT);
-            const uint32_t* lo_cr =
-                piv + ((uint64_t)R.piv_base[r] + j0) * R.n_runs;
-            const uint32_t* hi_cr =
-                (j1 + 1 < n_piv_r)
-                    ? piv + ((uint64_t)R.piv_base[r] + j1 + 1) * R.n_runs
-                    : nullptr;
-
-            for (int r2 = 0; r2 < R.n_runs; r2++) {
-                if (r2 == r) continue;
-                uint64_t lo = lo_cr[r2];
-                uint64_t hi = hi_cr ? (uint64_t)hi_cr[r2] : R.count[r2];
-                bool eq;
-                uint64_t cr =
-                    lower_rank(R, pfx, r2, e[0], epfx[0], r, lo, hi, eq);
-                rank[0] += cr;
-                winner[0] &= !eq;
-                #pragma unroll
-                for (int b = 1; b < RANK_BATCH; b++) {
-                    if (b >= nb) break;
-                    cr = advance_rank(R, pfx, r2, e[b], epfx[b], r, cr, hi,
-                                      eq);
-                    rank[b] += cr;
-                    winner[b] &= !eq;
-                }
-            }
-
-            #pragma unroll
-            for (int b = 0; b < RANK_BATCH; b++) {
-                if (b >= nb) break;
-                uint64_t dlen = (uint64_t)e[b].full_size - 32 - e[b].klen;
-                bool keep =
-                    winner[b] && (keep_tombstones || dlen != 0);
-                EntryMeta m;
-                m.src = ((uint64_t)r << 48) | e[b].off;
-                m.key_size = e[b].key_size;
-                m.full_size = e[b].full_size;
-                meta[rank[b]] = m;
-                sizes[rank[b]] = keep ? e[b].full_size : 0;
-                flags[rank[b]] = keep ? 1u : 0u;
-            }
-            gseg += nb;
-        }
+        uint64_t dlen = (uint64_t)e.full_size - 32 - e.klen;
+        bool keep = winner && (keep_tombstones || dlen != 0);
+        EntryMeta m;
+        m.src = ((uint64_t)r << 48) | e.off;
+        m.key_size = e.key_size;
+        m.full_size = e.full_size;
+        meta[rank] = m;
+        sizes[rank] = keep ? e.full_size : 0;
+        flags[rank] = keep ? 1u : 0u;
     }
 }
 
@@ -617,7 +613,11 @@ struct dbeel_gpu_job {
     size_t scantmp_bytes = 0;
     uint32_t* d_err = nullptr;
     uint64_t* d_pfx = nullptr;   /* dense big-endian key prefixes       */
-    uint32_t* d_piv = nullptr;   /* pivot crossranks                    */
+    void* d_aux = nullptr;       /* dense 64B search records            */
+    uint32_t* d_cr = nullptr;    /* per-pair crossranks, column-major   */
+    void* d_pairs = nullptr;     /* PairDesc table                      */
+    uint32_t n_pairs = 0;
+    uint64_t total_chunks = 0;
     uint32_t* d_winp0 = nullptr; /* copy window -> first survivor       */
     uint64_t total_entries = 0;
     uint64_t total_data_bytes = 0;
@@ -726,11 +726,9 @@ extern "C" int dbeel_gpu_job_create(const dbeel_run_view* runs, size_t n_runs,
     JOB_CHECK(hipMalloc(&job->d_outdata, total_data ? total_data : 16));
     JOB_CHECK(hipMalloc(&job->d_err, 2 * sizeof(uint32_t)));
     JOB_CHECK(hipMalloc(&job->d_pfx, n * sizeof(uint64_t)));
-    uint64_t n_piv_total = 0;
-    for (size_t r = 0; r < n_runs; r++)
-        n_piv_total += (runs[r].index_len / 16 + PIV_STRIDE - 1) >> PIV_SHIFT;
-    JOB_CHECK(hipMalloc(&job->d_piv,
-                        (n_piv_total ? n_piv_total : 1) * n_runs * 4));
+    JOB_CHECK(hipMalloc(&job->d_aux, n * 64));
+    JOB_CHECK(hipMalloc(&job->d_cr,
+                        (n_runs > 1 ? (n_runs - 1) * n : 1) * 4));
     JOB_CHECK(hipMalloc(&job->d_winp0,
                         (total_data / COPY_WINDOW + 2) * sizeof(uint32_t)));
 
@@ -770,12 +768,45 @@ extern "C" int dbeel_gpu_job_create(const dbeel_run_view* runs, size_t n_runs,
         D.entry_base[r] = base;
         base += D.count[r];
     }
-    uint32_t pbase = 0;
-    for (size_t r = 0; r < n_runs; r++) {
-        D.piv_base[r] = pbase;
-        pbase += (uint32_t)((D.count[r] + PIV_STRIDE - 1) >> PIV_SHIFT);
+    /* merge-path pair table: one entry per unordered run pair with work */
+    {
+        struct HostPair { uint32_t a, b; uint64_t chunk_base; };
+        size_t max_pairs = n_runs * (n_runs - 1) / 2 + 1;
+        HostPair* hp = (HostPair*)malloc(max_pairs * sizeof(HostPair));
+        if (!hp) {
+            dbeel_gpu_job_destroy(job);
+            return DBEEL_ERR_OOM;
+        }
+        uint64_t cbase = 0;
+        uint32_t np = 0;
+        for (uint32_t a = 0; a < n_runs; a++)
+            for (uint32_t b = a + 1; b < n_runs; b++) {
+                uint64_t len = D.count[a] + D.count[b];
+                if (!len) continue;
+                hp[np].a = a;
+                hp[np].b = b;
+                hp[np].chunk_base = cbase;
+                cbase += (len + CORANK_CHUNK - 1) / CORANK_CHUNK;
+                np++;
+            }
+        job->n_pairs = np;
+        job->total_chunks = cbase;
+        if (np) {
+            hipError_t _e = hipMalloc(&job->d_pairs, np * sizeof(PairDesc));
+            if (_e == hipSuccess)
+                _e = hipMemcpy(job->d_pairs, hp, np * sizeof(PairDesc),
+                               hipMemcpyHostToDevice);
+            free(hp);
+            if (_e != hipSuccess) {
+                set_err("pair table upload failed: %s",
+                        hipGetErrorString(_e));
+                dbeel_gpu_job_destroy(job);
+                return DBEEL_ERR_HIP;
+            }
+        } else {
+            free(hp);
+        }
     }
-    D.n_pivots = pbase;
     JOB_CHECK(hipEventRecord(job->ev[1], job->stream));
     JOB_CHECK(hipStreamSynchronize(job->stream));
     float ms = 0;
@@ -802,7 +833,9 @@ extern "C" void dbeel_gpu_job_destroy(dbeel_gpu_job* job) {
     hipFree(job->d_scantmp);
     hipFree(job->d_err);
     hipFree(job->d_pfx);
-    hipFree(job->d_piv);
+    hipFree(job->d_aux);
+    hipFree(job->d_cr);
+    hipFree(job->d_pairs);
     hipFree(job->d_winp0);
     for (int i = 0; i < 8; i++)
         if (job->ev[i]) hipEventDestroy(job->ev[i]);
@@ -838,19 +871,22 @@ extern "C" int dbeel_gpu_job_run(dbeel_gpu_job* job, int keep_tombstones,
     if (n) {
         uint32_t grid = pick_grid(n, 256);
         hipLaunchKernelGGL(k_prepare, dim3(grid), dim3(256), 0, s, job->desc,
-                           job->d_pfx, job->d_err);
-        if (job->desc.n_pivots)
-            hipLaunchKernelGGL(k_pivots,
-                               dim3(pick_grid(job->desc.n_pivots, 256)),
-                               dim3(256), 0, s, job->desc, job->d_pfx,
-                               job->d_piv);
+                           job->d_pfx, (Aux*)job->d_aux, job->d_err);
     }
     HIP_CHECK(hipEventRecord(job->ev[6], s));
     if (n) {
         uint32_t grid = pick_grid(n, 256);
-        hipLaunchKernelGGL(k_rank, dim3(grid), dim3(256), 0, s, job->desc,
-                           job->d_pfx, job->d_piv, job->d_meta, job->d_sizes,
-                           job->d_flags, keep_tombstones);
+        if (job->n_pairs)
+            hipLaunchKernelGGL(k_corank,
+                               dim3(pick_grid(job->total_chunks, 256)),
+                               dim3(256), 0, s, job->desc, job->d_pfx,
+                               (const Aux*)job->d_aux,
+                               (const PairDesc*)job->d_pairs, job->n_pairs,
+                               job->total_chunks, job->d_cr);
+        hipLaunchKernelGGL(k_rankreduce, dim3(grid), dim3(256), 0, s,
+                           job->desc, (const Aux*)job->d_aux, job->d_cr,
+                           job->d_meta, job->d_sizes, job->d_flags,
+                           keep_tombstones, job->d_err);
     }
     HIP_CHECK(hipEventRecord(job->ev[1], s));
     if (n) {
