@@ -78,13 +78,6 @@ struct RunsDesc {
     uint64_t total;                 /* sum of count                     */
 };
 
-/* 16 B/entry scratch record, scattered by global rank in k_rank. */
-struct EntryMeta {
-    uint64_t src;        /* run (high 16 bits) | data offset (low 48)  */
-    uint32_t key_size;   /* 8 + key_len (index record field)           */
-    uint32_t full_size;
-};
-
 struct EView {
     const uint8_t* raw;
     const uint8_t* key;
@@ -301,125 +294,198 @@ __global__ void k_prepare(RunsDesc R, uint64_t* pfx, Aux* aux,
 }
 
 /* ------------------------------------------------------------------ */
-/* Crossranks by merge-path co-ranking                                 */
+/* Crossranks by block-cooperative merge-path co-ranking               */
 /*                                                                     */
 /* For every run pair (a, b), the merged sequence (full order: key,    */
 /* timestamp, run index — a strict total order, so no merge ties) is   */
-/* partitioned into CORANK_CHUNK-wide chunks by diagonal binary        */
-/* search; each chunk is walked linearly, recording for every consumed */
-/* entry its crossrank into the opposite run (= the opposite cursor)   */
-/* and whether the opposite run's next entry carries the same key      */
-/* (then it is later in the order and supersedes it — the newest-wins  */
-/* dedup of lsm_tree.rs:1041-1046). Work is (k-1)*N sequential,        */
-/* cache-resident compares — not N*(k-1)*log2(N) random probes.        */
+/* partitioned into CORANK_BLOCK_POS-wide windows by diagonal binary   */
+/* search. Each 256-thread block stages the two pfx segments of its    */
+/* window into LDS with coalesced loads, sub-partitions the window     */
+/* 16 merged positions per thread (diagonal search inside LDS), and    */
+/* walks linearly, recording for every consumed entry its crossrank    */
+/* into the opposite run (= the opposite cursor) and whether the       */
+/* opposite run's next entry carries the same key (then it is later    */
+/* in the order and supersedes it — the newest-wins dedup of           */
+/* lsm_tree.rs:1041-1046). Work is (k-1)*N sequential LDS compares;    */
+/* global traffic is one coalesced pass over pfx plus rare aux reads   */
+/* on prefix ties (duplicate keys).                                    */
 /*                                                                     */
 /* cr layout: column-major, cr[s * total + g] = crossrank of entry g   */
 /* into the s-th OTHER run of g's run (s = r2 < r ? r2 : r2-1), with   */
 /* bit 31 = superseded flag. Every slot is written exactly once.       */
 /* ------------------------------------------------------------------ */
 
-#define CORANK_CHUNK 128
+#define CORANK_BLOCK 256
+#define CORANK_STEPS 16
+#define CORANK_BLOCK_POS (CORANK_BLOCK * CORANK_STEPS) /* 4096 */
 #define CR_LOSER 0x80000000u
 #define CR_MASK 0x7FFFFFFFu
 
 struct PairDesc {
     uint32_t a, b;
-    uint64_t chunk_base; /* exclusive prefix sum of per-pair chunks */
+    uint64_t chunk_base; /* exclusive prefix sum of per-pair windows */
 };
 
-/* strict full-order compare of a[ia] vs b[ib] on dense arrays */
-__device__ __forceinline__ int cmp_pair(const RunsDesc& R,
-                                        const uint64_t* pfx, const Aux* aux,
-                                        int ra, uint64_t ia, int rb,
-                                        uint64_t ib) {
-    uint64_t pa = pfx[R.entry_base[ra] + ia];
-    uint64_t pb = pfx[R.entry_base[rb] + ib];
-    if (pa != pb) return pa < pb ? -1 : 1;
+/* strict full-order compare of a[ia] vs b[ib]; pfx values supplied by
+ * the caller (staged in LDS or read globally) */
+__device__ __forceinline__ int cmp_tail_aux(const RunsDesc& R,
+                                            const Aux* aux, int ra,
+                                            uint64_t ia, int rb,
+                                            uint64_t ib) {
     return cmp_aux_full(R, aux, ra, ia, rb, ib);
 }
 
-__device__ __forceinline__ bool keys_eq_pair(const RunsDesc& R,
-                                             const uint64_t* pfx,
-                                             const Aux* aux, int ra,
-                                             uint64_t ia, int rb,
-                                             uint64_t ib) {
-    if (pfx[R.entry_base[ra] + ia] != pfx[R.entry_base[rb] + ib])
-        return false;
-    return cmp_keys_aux(R, aux, ra, ia, rb, ib) == 0;
-}
+__global__ __launch_bounds__(CORANK_BLOCK) void k_corank(
+    RunsDesc R, const uint64_t* pfx, const Aux* aux, const PairDesc* pairs,
+    uint32_t n_pairs, uint64_t total_chunks, uint32_t* cr) {
+    __shared__ uint64_t s_pfx[CORANK_BLOCK_POS + 2];
+    __shared__ uint64_t s_bounds[4]; /* iaS, ibS, iaE, ibE */
 
-__global__ void k_corank(RunsDesc R, const uint64_t* pfx, const Aux* aux,
-                         const PairDesc* pairs, uint32_t n_pairs,
-                         uint64_t total_chunks, uint32_t* cr) {
-    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
-    for (uint64_t t = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
-         t < total_chunks; t += stride) {
-        /* largest pair with chunk_base <= t */
-        uint32_t lo = 0, hi = n_pairs;
-        while (lo < hi) {
-            uint32_t mid = (lo + hi) >> 1;
+    for (uint64_t t = blockIdx.x; t < total_chunks; t += gridDim.x) {
+        /* largest pair with chunk_base <= t (uniform across the block) */
+        uint32_t plo = 0, phi = n_pairs;
+        while (plo < phi) {
+            uint32_t mid = (plo + phi) >> 1;
             if (pairs[mid].chunk_base <= t)
-                lo = mid + 1;
+                plo = mid + 1;
             else
-                hi = mid;
+                phi = mid;
         }
-        const PairDesc P = pairs[lo - 1];
+        const PairDesc P = pairs[plo - 1];
         const int a = (int)P.a, b = (int)P.b;
         const uint64_t Na = R.count[a], Nb = R.count[b];
-        uint64_t diag = (t - P.chunk_base) * CORANK_CHUNK;
-        uint64_t end = diag + CORANK_CHUNK;
-        if (end > Na + Nb) end = Na + Nb;
+        const uint64_t* gpa = pfx + R.entry_base[a];
+        const uint64_t* gpb = pfx + R.entry_base[b];
+        uint64_t diag0 = (t - P.chunk_base) * CORANK_BLOCK_POS;
+        uint64_t diag1 = diag0 + CORANK_BLOCK_POS;
+        if (diag1 > Na + Nb) diag1 = Na + Nb;
 
-        /* diagonal search: ia = # of a-entries among the first `diag`
-         * merged entries */
-        uint64_t slo = diag > Nb ? diag - Nb : 0;
-        uint64_t shi = diag < Na ? diag : Na;
-        while (slo < shi) {
-            uint64_t mid = (slo + shi) >> 1;
-            if (cmp_pair(R, pfx, aux, a, mid, b, diag - mid - 1) < 0)
-                slo = mid + 1;
-            else
-                shi = mid;
+        /* window corners by global diagonal search (threads 0 and 1) */
+        if (threadIdx.x < 2) {
+            uint64_t diag = threadIdx.x ? diag1 : diag0;
+            uint64_t slo = diag > Nb ? diag - Nb : 0;
+            uint64_t shi = diag < Na ? diag : Na;
+            while (slo < shi) {
+                uint64_t mid = (slo + shi) >> 1;
+                uint64_t pa = gpa[mid], pb = gpb[diag - mid - 1];
+                int c = (pa != pb)
+                            ? (pa < pb ? -1 : 1)
+                            : cmp_tail_aux(R, aux, a, mid, b, diag - mid - 1);
+                if (c < 0)
+                    slo = mid + 1;
+                else
+                    shi = mid;
+            }
+            s_bounds[threadIdx.x * 2] = slo;            /* ia */
+            s_bounds[threadIdx.x * 2 + 1] = diag - slo; /* ib */
         }
-        uint64_t ia = slo, ib = diag - slo;
+        __syncthreads();
+        const uint64_t iaS = s_bounds[0], ibS = s_bounds[1];
+        const uint64_t iaE = s_bounds[2], ibE = s_bounds[3];
+        const uint32_t lenA = (uint32_t)(iaE - iaS);
+        const uint32_t lenB = (uint32_t)(ibE - ibS);
 
-        uint32_t sa = (uint32_t)(b < a ? b : b - 1); /* slot of b in a */
-        uint32_t sb = (uint32_t)(a < b ? a : a - 1); /* slot of a in b */
-        uint32_t* cra = cr + (uint64_t)sa * R.total + R.entry_base[a];
-        uint32_t* crb = cr + (uint64_t)sb * R.total + R.entry_base[b];
+        /* stage both pfx segments (coalesced); sA = s_pfx[0..lenA),
+         * sB = s_pfx[lenA..lenA+lenB) */
+        for (uint32_t u = threadIdx.x; u < lenA; u += CORANK_BLOCK)
+            s_pfx[u] = gpa[iaS + u];
+        for (uint32_t u = threadIdx.x; u < lenB; u += CORANK_BLOCK)
+            s_pfx[lenA + u] = gpb[ibS + u];
+        __syncthreads();
+        const uint64_t* sA = s_pfx;
+        const uint64_t* sB = s_pfx + lenA;
 
-        for (uint64_t pos = diag; pos < end; pos++) {
-            bool take_a;
-            if (ia >= Na)
-                take_a = false;
-            else if (ib >= Nb)
-                take_a = true;
-            else
-                take_a = cmp_pair(R, pfx, aux, a, ia, b, ib) < 0;
-            if (take_a) {
-                uint32_t v = (uint32_t)ib;
-                if (ib < Nb && keys_eq_pair(R, pfx, aux, a, ia, b, ib))
-                    v |= CR_LOSER;
-                cra[ia] = v;
-                ia++;
-            } else {
-                uint32_t v = (uint32_t)ia;
-                if (ia < Na && keys_eq_pair(R, pfx, aux, b, ib, a, ia))
-                    v |= CR_LOSER;
-                crb[ib] = v;
-                ib++;
+        /* per-thread sub-window: local diagonal search inside LDS */
+        uint32_t L = lenA + lenB;
+        uint32_t d = threadIdx.x * CORANK_STEPS;
+        if (d < L) {
+            uint32_t slo = d > lenB ? d - lenB : 0;
+            uint32_t shi = d < lenA ? d : lenA;
+            while (slo < shi) {
+                uint32_t mid = (slo + shi) >> 1;
+                uint64_t pa = sA[mid], pb = sB[d - mid - 1];
+                int c = (pa != pb) ? (pa < pb ? -1 : 1)
+                                   : cmp_tail_aux(R, aux, a, iaS + mid, b,
+                                                  ibS + d - mid - 1);
+                if (c < 0)
+                    slo = mid + 1;
+                else
+                    shi = mid;
+            }
+            uint32_t ja = slo, jb = d - slo;
+            uint32_t dend = d + CORANK_STEPS;
+            if (dend > L) dend = L;
+
+            uint32_t sa = (uint32_t)(b < a ? b : b - 1); /* slot of b in a */
+            uint32_t sb = (uint32_t)(a < b ? a : a - 1); /* slot of a in b */
+            uint32_t* cra = cr + (uint64_t)sa * R.total + R.entry_base[a];
+            uint32_t* crb = cr + (uint64_t)sb * R.total + R.entry_base[b];
+
+            for (uint32_t pos = d; pos < dend; pos++) {
+                bool take_a;
+                if (ja >= lenA)
+                    take_a = false;
+                else if (jb >= lenB)
+                    take_a = true;
+                else {
+                    uint64_t pa = sA[ja], pb = sB[jb];
+                    take_a = (pa != pb)
+                                 ? (pa < pb)
+                                 : (cmp_tail_aux(R, aux, a, iaS + ja, b,
+                                                 ibS + jb) < 0);
+                }
+                if (take_a) {
+                    uint64_t gib = ibS + jb;
+                    uint32_t v = (uint32_t)gib;
+                    /* next b entry: staged, or first beyond the window */
+                    if (gib < Nb) {
+                        uint64_t pb =
+                            (jb < lenB) ? sB[jb] : gpb[gib];
+                        if (pb == sA[ja] &&
+                            cmp_keys_aux(R, aux, a, iaS + ja, b, gib) == 0)
+                            v |= CR_LOSER;
+                    }
+                    cra[iaS + ja] = v;
+                    ja++;
+                } else {
+                    uint64_t gia = iaS + ja;
+                    uint32_t v = (uint32_t)gia;
+                    if (gia < Na) {
+                        uint64_t pa =
+                            (ja < lenA) ? sA[ja] : gpa[gia];
+                        if (pa == sB[jb] &&
+                            cmp_keys_aux(R, aux, b, ibS + jb, a, gia) == 0)
+                            v |= CR_LOSER;
+                    }
+                    crb[ibS + jb] = v;
+                    jb++;
+                }
             }
         }
+        __syncthreads();
     }
 }
+
+/* Rank-indexed scratch record: one 32-B scattered write per entry
+ * (vs three separate arrays = 3 scattered lines). size == 0 marks a
+ * dropped entry; the survivor scans read it through strided transform
+ * iterators. */
+struct RankRec {
+    uint64_t src;      /* run (high 16 bits) | data offset (low 48) */
+    uint32_t key_size; /* 8 + key_len (index record field)          */
+    uint32_t full_size;
+    uint64_t size;     /* full_size if kept, else 0                 */
+    uint64_t pad;
+};
+static_assert(sizeof(RankRec) == 32, "rank record must be 32B");
 
 /* Reduce per-pair crossranks to the global rank, apply the winner /
  * tombstone rules, and emit the rank-indexed scratch records. Also checks
  * each run is strictly sorted by key (flush invariant,
- * lsm_tree.rs:925-946) on the dense aux records. */
-__global__ void k_rankreduce(RunsDesc R, const Aux* aux, const uint32_t* cr,
-                             EntryMeta* meta, uint64_t* sizes,
-                             uint32_t* flags, int keep_tombstones,
+ * lsm_tree.rs:925-946) on the dense pfx/aux arrays. */
+__global__ void k_rankreduce(RunsDesc R, const uint64_t* pfx,
+                             const Aux* aux, const uint32_t* cr,
+                             RankRec* rrec, int keep_tombstones,
                              uint32_t* err) {
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
     for (uint64_t g = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
@@ -431,14 +497,16 @@ __global__ void k_rankreduce(RunsDesc R, const Aux* aux, const uint32_t* cr,
         if (!load_entry(R, r, i, e)) {
             /* k_prepare has flagged this input; keep memory safe and park
              * the entry at its local slot (results will be discarded) */
-            meta[g] = EntryMeta{0, 8, 32};
-            sizes[g] = 0;
-            flags[g] = 0;
+            RankRec z = {0, 8, 32, 0, 0};
+            rrec[g] = z;
             continue;
         }
-        if (i + 1 < R.count[r] &&
-            cmp_keys_aux(R, aux, r, i, r, i + 1) >= 0)
-            atomicOr(err, DERR_UNSORTED);
+        if (i + 1 < R.count[r]) {
+            uint64_t p0 = pfx[g], p1 = pfx[g + 1];
+            if (p0 > p1 ||
+                (p0 == p1 && cmp_keys_aux(R, aux, r, i, r, i + 1) >= 0))
+                atomicOr(err, DERR_UNSORTED);
+        }
 
         uint64_t rank = i;
         bool winner = true;
@@ -449,31 +517,31 @@ __global__ void k_rankreduce(RunsDesc R, const Aux* aux, const uint32_t* cr,
         }
         uint64_t dlen = (uint64_t)e.full_size - 32 - e.klen;
         bool keep = winner && (keep_tombstones || dlen != 0);
-        EntryMeta m;
+        RankRec m;
         m.src = ((uint64_t)r << 48) | e.off;
         m.key_size = e.key_size;
         m.full_size = e.full_size;
-        meta[rank] = m;
-        sizes[rank] = keep ? e.full_size : 0;
-        flags[rank] = keep ? 1u : 0u;
+        m.size = keep ? e.full_size : 0;
+        m.pad = 0;
+        rrec[rank] = m;
     }
 }
 
 /* Output index records are the input format: offset u64 | key_size u32 |
  * full_size u32 (entry_writer.rs:79-87, offsets recomputed from 0).
  * src_map gets the ABSOLUTE device address of each survivor's bytes. */
-__global__ void k_emit(RunsDesc R, const EntryMeta* meta,
-                       const uint64_t* sizes, const uint64_t* dst_off,
-                       const uint32_t* pos, uint64_t total,
-                       uint8_t* out_index, uint64_t* src_map) {
+__global__ void k_emit(RunsDesc R, const RankRec* rrec,
+                       const uint64_t* dst_off, const uint32_t* pos,
+                       uint64_t total, uint8_t* out_index,
+                       uint64_t* src_map) {
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
     for (uint64_t g = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
          g < total; g += stride) {
-        if (!sizes[g]) continue;
+        RankRec m = rrec[g];
+        if (!m.size) continue;
         uint32_t p = pos[g];
         uint8_t* rec = out_index + (uint64_t)p * 16;
         uint64_t off = dst_off[g];
-        EntryMeta m = meta[g];
         __builtin_memcpy(rec, &off, 8);
         __builtin_memcpy(rec + 8, &m.key_size, 4);
         __builtin_memcpy(rec + 12, &m.full_size, 4);
@@ -481,6 +549,16 @@ __global__ void k_emit(RunsDesc R, const EntryMeta* meta,
             (uint64_t)(R.data[m.src >> 48] + (m.src & 0xFFFFFFFFFFFFull));
     }
 }
+
+/* transform-iterator functors for the survivor scans */
+struct RankRecSize {
+    __device__ uint64_t operator()(const RankRec& r) const { return r.size; }
+};
+struct RankRecFlag {
+    __device__ uint32_t operator()(const RankRec& r) const {
+        return r.size ? 1u : 0u;
+    }
+};
 
 /* Balanced verbatim copy: a 16-KiB destination window per 256-thread
  * block, 4 x 16-B granules per thread (independent loads/stores for ILP;
@@ -601,9 +679,7 @@ struct dbeel_gpu_job {
     hipEvent_t ev[8] = {};
     RunsDesc desc{};
     uint8_t* d_input = nullptr; /* one slab: all run data+index            */
-    EntryMeta* d_meta = nullptr;
-    uint64_t* d_sizes = nullptr;
-    uint32_t* d_flags = nullptr;
+    RankRec* d_rank = nullptr;
     uint64_t* d_dstoff = nullptr;
     uint32_t* d_pos = nullptr;
     uint8_t* d_outindex = nullptr;
@@ -716,9 +792,7 @@ extern "C" int dbeel_gpu_job_create(const dbeel_run_view* runs, size_t n_runs,
 
     JOB_CHECK(hipMalloc(&job->d_input, input_bytes ? input_bytes : 16));
     uint64_t n = total ? total : 1;
-    JOB_CHECK(hipMalloc(&job->d_meta, n * sizeof(EntryMeta)));
-    JOB_CHECK(hipMalloc(&job->d_sizes, n * sizeof(uint64_t)));
-    JOB_CHECK(hipMalloc(&job->d_flags, n * sizeof(uint32_t)));
+    JOB_CHECK(hipMalloc(&job->d_rank, n * sizeof(RankRec)));
     JOB_CHECK(hipMalloc(&job->d_dstoff, n * sizeof(uint64_t)));
     JOB_CHECK(hipMalloc(&job->d_pos, n * sizeof(uint32_t)));
     JOB_CHECK(hipMalloc(&job->d_outindex, n * 16));
@@ -733,12 +807,16 @@ extern "C" int dbeel_gpu_job_create(const dbeel_run_view* runs, size_t n_runs,
                         (total_data / COPY_WINDOW + 2) * sizeof(uint32_t)));
 
     size_t t1 = 0, t2 = 0;
-    rocprim::exclusive_scan(nullptr, t1, job->d_sizes, job->d_dstoff,
-                            (uint64_t)0, n, rocprim::plus<uint64_t>(),
-                            job->stream);
-    rocprim::exclusive_scan(nullptr, t2, job->d_flags, job->d_pos,
-                            (uint32_t)0, n, rocprim::plus<uint32_t>(),
-                            job->stream);
+    rocprim::exclusive_scan(nullptr, t1,
+                            rocprim::make_transform_iterator(job->d_rank,
+                                                             RankRecSize{}),
+                            job->d_dstoff, (uint64_t)0, n,
+                            rocprim::plus<uint64_t>(), job->stream);
+    rocprim::exclusive_scan(nullptr, t2,
+                            rocprim::make_transform_iterator(job->d_rank,
+                                                             RankRecFlag{}),
+                            job->d_pos, (uint32_t)0, n,
+                            rocprim::plus<uint32_t>(), job->stream);
     job->scantmp_bytes = t1 > t2 ? t1 : t2;
     JOB_CHECK(hipMalloc(&job->d_scantmp, job->scantmp_bytes));
 
@@ -786,7 +864,7 @@ extern "C" int dbeel_gpu_job_create(const dbeel_run_view* runs, size_t n_runs,
                 hp[np].a = a;
                 hp[np].b = b;
                 hp[np].chunk_base = cbase;
-                cbase += (len + CORANK_CHUNK - 1) / CORANK_CHUNK;
+                cbase += (len + CORANK_BLOCK_POS - 1) / CORANK_BLOCK_POS;
                 np++;
             }
         job->n_pairs = np;
@@ -822,9 +900,7 @@ extern "C" void dbeel_gpu_job_destroy(dbeel_gpu_job* job) {
     if (!job) return;
     if (job->device >= 0) hipSetDevice(job->device);
     hipFree(job->d_input);
-    hipFree(job->d_meta);
-    hipFree(job->d_sizes);
-    hipFree(job->d_flags);
+    hipFree(job->d_rank);
     hipFree(job->d_dstoff);
     hipFree(job->d_pos);
     hipFree(job->d_outindex);
@@ -876,46 +952,49 @@ extern "C" int dbeel_gpu_job_run(dbeel_gpu_job* job, int keep_tombstones,
     HIP_CHECK(hipEventRecord(job->ev[6], s));
     if (n) {
         uint32_t grid = pick_grid(n, 256);
-        if (job->n_pairs)
-            hipLaunchKernelGGL(k_corank,
-                               dim3(pick_grid(job->total_chunks, 256)),
-                               dim3(256), 0, s, job->desc, job->d_pfx,
-                               (const Aux*)job->d_aux,
+        if (job->n_pairs) {
+            uint64_t cgrid = job->total_chunks;
+            if (cgrid > 16384) cgrid = 16384;
+            hipLaunchKernelGGL(k_corank, dim3((uint32_t)cgrid),
+                               dim3(CORANK_BLOCK), 0, s, job->desc,
+                               job->d_pfx, (const Aux*)job->d_aux,
                                (const PairDesc*)job->d_pairs, job->n_pairs,
                                job->total_chunks, job->d_cr);
+        }
         hipLaunchKernelGGL(k_rankreduce, dim3(grid), dim3(256), 0, s,
-                           job->desc, (const Aux*)job->d_aux, job->d_cr,
-                           job->d_meta, job->d_sizes, job->d_flags,
-                           keep_tombstones, job->d_err);
+                           job->desc, job->d_pfx, (const Aux*)job->d_aux,
+                           job->d_cr, job->d_rank, keep_tombstones,
+                           job->d_err);
     }
     HIP_CHECK(hipEventRecord(job->ev[1], s));
     if (n) {
         size_t tmp = job->scantmp_bytes;
-        rocprim::exclusive_scan(job->d_scantmp, tmp, job->d_sizes,
-                                job->d_dstoff, (uint64_t)0, n,
-                                rocprim::plus<uint64_t>(), s);
+        (void)rocprim::exclusive_scan(
+            job->d_scantmp, tmp,
+            rocprim::make_transform_iterator(job->d_rank, RankRecSize{}),
+            job->d_dstoff, (uint64_t)0, n, rocprim::plus<uint64_t>(), s);
         tmp = job->scantmp_bytes;
-        rocprim::exclusive_scan(job->d_scantmp, tmp, job->d_flags, job->d_pos,
-                                (uint32_t)0, n, rocprim::plus<uint32_t>(), s);
+        (void)rocprim::exclusive_scan(
+            job->d_scantmp, tmp,
+            rocprim::make_transform_iterator(job->d_rank, RankRecFlag{}),
+            job->d_pos, (uint32_t)0, n, rocprim::plus<uint32_t>(), s);
     }
     HIP_CHECK(hipEventRecord(job->ev[2], s));
     if (n) {
         uint32_t grid = pick_grid(n, 256);
         hipLaunchKernelGGL(k_emit, dim3(grid), dim3(256), 0, s, job->desc,
-                           job->d_meta, job->d_sizes, job->d_dstoff,
-                           job->d_pos, n, job->d_outindex, job->d_srcmap);
+                           job->d_rank, job->d_dstoff, job->d_pos, n,
+                           job->d_outindex, job->d_srcmap);
     }
     HIP_CHECK(hipEventRecord(job->ev[3], s));
 
     /* Need totals on host to size/launch the copy; one small sync. */
     uint64_t last_size = 0, last_off = 0;
-    uint32_t last_flag = 0, last_pos = 0, err = 0;
+    uint32_t last_pos = 0, err = 0;
     if (n) {
-        HIP_CHECK(hipMemcpyAsync(&last_size, job->d_sizes + (n - 1), 8,
+        HIP_CHECK(hipMemcpyAsync(&last_size, &job->d_rank[n - 1].size, 8,
                                  hipMemcpyDeviceToHost, s));
         HIP_CHECK(hipMemcpyAsync(&last_off, job->d_dstoff + (n - 1), 8,
-                                 hipMemcpyDeviceToHost, s));
-        HIP_CHECK(hipMemcpyAsync(&last_flag, job->d_flags + (n - 1), 4,
                                  hipMemcpyDeviceToHost, s));
         HIP_CHECK(hipMemcpyAsync(&last_pos, job->d_pos + (n - 1), 4,
                                  hipMemcpyDeviceToHost, s));
@@ -930,7 +1009,7 @@ extern "C" int dbeel_gpu_job_run(dbeel_gpu_job* job, int keep_tombstones,
         return DBEEL_ERR_CORRUPT;
     }
     uint64_t total_out = last_off + last_size;
-    uint64_t n_surv = (uint64_t)last_pos + last_flag;
+    uint64_t n_surv = (uint64_t)last_pos + (last_size ? 1 : 0);
 
     HIP_CHECK(hipEventRecord(job->ev[4], s));
     if (total_out) {
