@@ -282,9 +282,19 @@ __global__ void k_prepare(RunsDesc R, uint64_t* pfx, Aux* aux,
         load_ts(e, a.ts_lo, a.ts_hi);
         uint32_t nk = e.klen < AUX_KEY_BYTES ? (uint32_t)e.klen
                                              : AUX_KEY_BYTES;
+        /* 8-byte chunked key copy; reading up to 7 bytes past the key is
+         * safe (the 8-byte data_len field follows it inside the entry)
+         * and the over-read is masked off */
         #pragma unroll
-        for (uint32_t b = 0; b < AUX_KEY_BYTES; b++)
-            a.key[b] = (b < nk) ? e.key[b] : 0;
+        for (uint32_t j = 0; j < AUX_KEY_BYTES; j += 8) {
+            uint64_t v = 0;
+            if (j < nk) {
+                v = ld_u64(e.key + j);
+                if (j + 8 > nk)
+                    v &= (~0ull) >> (8 * (j + 8 - nk));
+            }
+            __builtin_memcpy(a.key + j, &v, 8);
+        }
         aux[g] = a;
         /* bincode field cross-check */
         if (ld_u64(e.raw) != e.klen ||
@@ -650,9 +660,13 @@ __global__ __launch_bounds__(COPY_BLOCK) void k_copy(
             uint8_t* dst = out_data + gpos;
             if (gpos + nbytes <= e_end) {
                 if (nbytes == 16) {
-                    uint4 v;
+                    typedef unsigned int v4u
+                        __attribute__((ext_vector_type(4)));
+                    v4u v;
                     __builtin_memcpy(&v, src, 16);
-                    *reinterpret_cast<uint4*>(dst) = v;
+                    /* streamed once, never re-read: keep L2 for sources */
+                    __builtin_nontemporal_store(
+                        v, reinterpret_cast<v4u*>(dst));
                 } else {
                     for (uint32_t b = 0; b < nbytes; b++) dst[b] = src[b];
                 }
