@@ -141,6 +141,18 @@ def main():
     }
     dom_ms = per_kernel[dom]
     achieved = (algo[dom] / 1e9) / (dom_ms / 1e3) if dom_ms > 0 else 0.0
+    # PMC-measured per-launch HBM traffic (rocprofv3 --pmc FETCH_SIZE /
+    # WRITE_SIZE in separate passes, gfx950 2x fetch correction calibrated
+    # against k_copy's known byte counts — see profiles/r01_traffic_cfg3.json)
+    traffic = None
+    pipe_traffic = None
+    tpath = os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                         "profiles", "r01_traffic_cfg3.json")
+    if (args.scale == 1.0 and os.path.exists(tpath)):
+        tj = json.load(open(tpath))
+        kmap = tj.get("kernels", {})
+        traffic = kmap.get(f"k_{dom}", {}).get("traffic_bytes")
+        pipe_traffic = sum(v["traffic_bytes"] for v in kmap.values())
     roofline = {
         "bound": "hbm",
         "kernel": f"k_{dom}",
@@ -148,7 +160,7 @@ def main():
         "peak": HBM_PEAK_GBPS,
         "unit": "GB/s",
         "frac": round(achieved / HBM_PEAK_GBPS, 4),
-        "traffic": None,  # PMC-measured per-launch HBM bytes: profiles/
+        "traffic": traffic,
     }
     # whole-pipeline roofline (SURVEY.md §8d algorithmic B / kernel time)
     B = input_bytes + out_bytes + out_index_bytes
@@ -159,7 +171,7 @@ def main():
         "peak": HBM_PEAK_GBPS,
         "unit": "GB/s",
         "frac": round(pipe_gbps / HBM_PEAK_GBPS, 4),
-        "traffic": None,
+        "traffic": pipe_traffic,
     }
 
     cpu_baseline = None
