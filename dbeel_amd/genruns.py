@@ -136,3 +136,56 @@ def make_config(name: str, seed: int = DEFAULT_SEED, scale: float = 1.0):
 
 def input_bytes(runs) -> int:
     return sum(len(d) + len(i) for d, i in runs)
+
+
+def make_runs_varkey(
+    n_runs: int,
+    entries_per_run: int,
+    value_size: int = 4096,
+    overlap_frac: float = 0.3,
+    tombstone_frac: float = 0.02,
+    seed: int = DEFAULT_SEED,
+):
+    """cfg5-shaped runs: variable-length msgpack str keys, zipf(1.1)
+    lengths clipped to 8..128 B total, 4 KiB values (BASELINE.json
+    configs[4]). Keys are raw msgpack `str8` encodings (0xd9 | len |
+    utf8-ish bytes) compared as raw bytes, exactly as dbeel compares them
+    (Entry::cmp on Vec<u8>, mod.rs:75-81). Python-loop builder — parity
+    test scale, not bench scale."""
+    from .format import Entry, build_run
+
+    rng = np.random.default_rng(seed)
+
+    def draw_keys(n):
+        keys = set()
+        while len(keys) < n:
+            L = int(np.clip(rng.zipf(1.1), 6, 126))
+            body = bytes(rng.integers(32, 127, L, dtype=np.uint8))
+            keys.add(bytes([0xD9, L]) + body)
+        return list(keys)
+
+    n_shared = int(entries_per_run * overlap_frac)
+    shared = draw_keys(n_shared * (n_runs // 2)) if n_shared else []
+    out = []
+    spos = 0
+    for r in range(n_runs):
+        own = draw_keys(entries_per_run - n_shared)
+        ks = own
+        if n_shared:
+            if r % 2 == 0 and spos + n_shared <= len(shared):
+                sh = shared[spos : spos + n_shared]
+            else:
+                sh = shared[max(0, spos - n_shared) : spos]
+            if r % 2 == 1:
+                spos += n_shared
+            ks = own + sh[: n_shared]
+        ks = sorted(set(ks))
+        ents = []
+        for i, k in enumerate(ks):
+            tomb = rng.random() < tombstone_frac
+            data = b"" if tomb else bytes(
+                rng.integers(0, 256, value_size, dtype=np.uint8)
+            )
+            ents.append(Entry(k, data, (r << 40) + i))
+        out.append(build_run(ents))
+    return out

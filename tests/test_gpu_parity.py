@@ -81,6 +81,21 @@ def test_ragged_keys_parity(engine):
         assert (gd, gi, gn) == (od, oi, on)
 
 
+def test_cfg5_varkey_16run_parity(engine):
+    """cfg5 shape (BASELINE configs[4]): 16-run merge of variable-length
+    zipf msgpack str keys + 4 KiB values, scaled to parity-test size."""
+    from dbeel_amd.genruns import make_runs_varkey
+
+    runs = make_runs_varkey(16, 800, value_size=4096, overlap_frac=0.3,
+                            tombstone_frac=0.05, seed=0x5E5)
+    for keep in (True, False):
+        gd, gi, gn = engine.compact(runs, keep_tombstones=keep, device=0)
+        od, oi, on = oracle.compact(runs, keep_tombstones=keep)
+        assert gn == on
+        assert gi == oi
+        assert gd == od
+
+
 def test_empty_inputs(engine):
     runs = [(b"", b""), (b"", b"")]
     gd, gi, gn = engine.compact(runs, keep_tombstones=True, device=0)
