@@ -28,14 +28,26 @@ sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 HBM_PEAK_GBPS = 8000.0  # MI355X HBM3E spec peak, GB/s (MI355X_MICROARCH.md)
 
 
-def make_workload(rank: int, scale: float):
+def make_workload(rank: int, scale: float, workload: str = "cfg3"):
+    """cfg3: one 8-run x 1 GiB job per GPU (the metric config).
+    cfg4: 8 independent 4-run x 256 MiB jobs per GPU (= BASELINE configs[3],
+    64 jobs over 8 GPUs, one dbeel shard each). Returns (job_list, cfg)."""
     from dbeel_amd.genruns import CONFIGS, make_runs
 
+    if workload == "cfg4":
+        cfg = dict(CONFIGS["cfg4_job"])
+        if scale != 1.0:
+            cfg["entries_per_run"] = max(64, int(cfg["entries_per_run"] * scale))
+        jobs = [
+            make_runs(seed=0xDBEE1 + 7919 * rank + 101 * j, **cfg)
+            for j in range(8)
+        ]
+        return jobs, cfg
     cfg = dict(CONFIGS["cfg3"])
     if scale != 1.0:
         cfg["entries_per_run"] = max(64, int(cfg["entries_per_run"] * scale))
     runs = make_runs(seed=0xDBEE1 + 7919 * rank, **cfg)
-    return runs, cfg
+    return [runs], cfg
 
 
 def rank_algorithmic_bytes(cfg, n_entries: int) -> int:
@@ -51,6 +63,9 @@ def main():
     ap.add_argument("--warmup", type=int, default=3)
     ap.add_argument("--scale", type=float, default=1.0,
                     help="scale entries_per_run (1.0 = full config 3)")
+    ap.add_argument("--workload", choices=["cfg3", "cfg4"], default="cfg3",
+                    help="cfg3 = 8-run x 1 GiB metric config; cfg4 = 8 "
+                         "independent 4-run x 256 MiB jobs per GPU")
     ap.add_argument("--keep-tombstones", action="store_true")
     ap.add_argument("--cpu-baseline-scale", type=float, default=0.25,
                     help="fraction of the workload timed on 1 host core")
@@ -74,17 +89,32 @@ def main():
     import dbeel_amd
 
     keep = bool(args.keep_tombstones)
-    runs, cfg = make_workload(rank, args.scale)
-    input_bytes = sum(d.nbytes + i.nbytes for d, i in runs)
-    n_entries = sum(i.nbytes // 16 for d, i in runs)
+    job_runs, cfg = make_workload(rank, args.scale, args.workload)
+    input_bytes = sum(
+        d.nbytes + i.nbytes for runs in job_runs for d, i in runs
+    )
+    n_entries = sum(i.nbytes // 16 for runs in job_runs for d, i in runs)
 
-    job = dbeel_amd.Job(runs, device=local_rank)
+    jobs = [dbeel_amd.Job(runs, device=local_rank) for runs in job_runs]
+
+    def run_all():
+        tb = te = 0
+        t_acc = None
+        for j in jobs:
+            ob, oe, t = j.run(keep)
+            tb += ob
+            te += oe
+            if t_acc is None:
+                t_acc = dict(t)
+            else:
+                for k in t_acc:
+                    t_acc[k] += t[k]
+        return tb, te, t_acc
 
     # Warmup (untimed)
     out_bytes = out_entries = 0
-    tim_acc = None
     for _ in range(args.warmup):
-        out_bytes, out_entries, _ = job.run(keep)
+        out_bytes, out_entries, _ = run_all()
 
     def barrier_sync():
         torch.cuda.synchronize(local_rank) if torch.cuda.is_available() else None
@@ -97,7 +127,7 @@ def main():
     t0 = time.perf_counter()
     tims = []
     for _ in range(args.steps):
-        out_bytes, out_entries, t = job.run(keep)
+        out_bytes, out_entries, t = run_all()
         tims.append(t)
         if dist:
             # the path's only collective: RCCL all-gather of emitted byte
@@ -148,7 +178,8 @@ def main():
     pipe_traffic = None
     tpath = os.path.join(os.path.dirname(os.path.abspath(__file__)),
                          "profiles", "r01_traffic_cfg3.json")
-    if (args.scale == 1.0 and os.path.exists(tpath)):
+    if (args.scale == 1.0 and args.workload == "cfg3"
+            and os.path.exists(tpath)):
         tj = json.load(open(tpath))
         kmap = tj.get("kernels", {})
         traffic = kmap.get(f"k_{dom}", {}).get("traffic_bytes")
@@ -179,10 +210,11 @@ def main():
         import oracle
 
         bs = args.cpu_baseline_scale * args.scale
-        cruns, _ = make_workload(0, bs)
-        cbytes = sum(d.nbytes + i.nbytes for d, i in cruns)
+        cjobs, _ = make_workload(0, bs, args.workload)
+        cbytes = sum(d.nbytes + i.nbytes for runs in cjobs for d, i in runs)
         c0 = time.perf_counter()
-        oracle.compact(cruns, keep_tombstones=keep)
+        for cruns in cjobs:
+            oracle.compact(cruns, keep_tombstones=keep)
         c1 = time.perf_counter()
         cpu_baseline = {
             "value": round((cbytes / 1e6) / (c1 - c0), 1),
@@ -190,7 +222,7 @@ def main():
             "cores": 1,
             "kind": "port",
             "sample": (
-                f"cfg3 shape at {bs:.2f} scale "
+                f"{args.workload} shape at {bs:.2f} scale "
                 f"({cbytes / 1e6:.0f} MB input, {c1 - c0:.1f}s on 1 core; "
                 "C oracle restatement — Rust/glommio unbuildable here, "
                 "BASELINE.md)"
@@ -212,7 +244,8 @@ def main():
             "dtype": "u8",
             "data": "synthetic",
             "config": {
-                "workload": "cfg3_8run_x_1GiB"
+                "workload": ("cfg3_8run_x_1GiB" if args.workload == "cfg3"
+                             else "cfg4_8jobs_4run_x_256MiB_per_gpu")
                 + (f"_scale{args.scale}" if args.scale != 1.0 else ""),
                 "n_runs": cfg["n_runs"],
                 "entries_per_run": cfg["entries_per_run"],
@@ -232,7 +265,8 @@ def main():
         }
         print(json.dumps(line))
 
-    job.close()
+    for j in jobs:
+        j.close()
     if dist:
         dist.destroy_process_group()
 

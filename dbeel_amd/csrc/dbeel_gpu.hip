@@ -684,6 +684,220 @@ __global__ __launch_bounds__(COPY_BLOCK) void k_copy(
 }
 
 /* ------------------------------------------------------------------ */
+/* Run encoder (the memtable-flush path)                              */
+/*                                                                    */
+/* Encodes an already-sorted (key, value, timestamp) stream into a    */
+/* run: the same bincode-fixint entry layout + 16-B index records as  */
+/* flush_memtable_to_disk / EntryWriter (lsm_tree.rs:925-946,         */
+/* entry_writer.rs:71-98). Wave-per-entry copy: lanes move 8-byte     */
+/* chunks of key/value bytes; the header fields and timestamp are     */
+/* written by lane 0.                                                 */
+/* ------------------------------------------------------------------ */
+
+static uint32_t pick_grid(uint64_t work_items, uint32_t block);
+
+__global__ void k_encode_sizes(uint64_t n, const uint64_t* key_off,
+                               const uint64_t* val_off, uint64_t* sizes) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         i < n; i += stride) {
+        sizes[i] = 32 + (key_off[i + 1] - key_off[i]) +
+                   (val_off[i + 1] - val_off[i]);
+    }
+}
+
+__global__ void k_encode_index(uint64_t n, const uint64_t* key_off,
+                               const uint64_t* dst_off,
+                               const uint64_t* sizes, uint8_t* out_index) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         i < n; i += stride) {
+        uint8_t* rec = out_index + i * 16;
+        uint64_t off = dst_off[i];
+        uint32_t key_size = (uint32_t)(8 + key_off[i + 1] - key_off[i]);
+        uint32_t full_size = (uint32_t)sizes[i];
+        __builtin_memcpy(rec, &off, 8);
+        __builtin_memcpy(rec + 8, &key_size, 4);
+        __builtin_memcpy(rec + 12, &full_size, 4);
+    }
+}
+
+__device__ __forceinline__ void wave_copy_bytes(uint8_t* dst,
+                                                const uint8_t* src,
+                                                uint64_t n, uint32_t lane) {
+    /* 8-byte chunks per lane, byte tail by lane 0 */
+    uint64_t chunks = n >> 3;
+    for (uint64_t c = lane; c < chunks; c += 64) {
+        uint64_t v;
+        __builtin_memcpy(&v, src + c * 8, 8);
+        __builtin_memcpy(dst + c * 8, &v, 8);
+    }
+    if (lane == 0)
+        for (uint64_t b = chunks * 8; b < n; b++) dst[b] = src[b];
+}
+
+__global__ void k_encode_data(uint64_t n, const uint8_t* keys,
+                              const uint64_t* key_off, const uint8_t* vals,
+                              const uint64_t* val_off, const uint8_t* ts,
+                              const uint64_t* dst_off, uint8_t* out_data) {
+    uint32_t lane = threadIdx.x & 63;
+    uint64_t wave = ((uint64_t)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+    uint64_t n_waves = ((uint64_t)gridDim.x * blockDim.x) >> 6;
+    for (uint64_t i = wave; i < n; i += n_waves) {
+        uint64_t klen = key_off[i + 1] - key_off[i];
+        uint64_t vlen = val_off[i + 1] - val_off[i];
+        uint8_t* dst = out_data + dst_off[i];
+        if (lane == 0) {
+            __builtin_memcpy(dst, &klen, 8);
+            __builtin_memcpy(dst + 8 + klen, &vlen, 8);
+            __builtin_memcpy(dst + 16 + klen + vlen, ts + i * 16, 16);
+        }
+        wave_copy_bytes(dst + 8, keys + key_off[i], klen, lane);
+        wave_copy_bytes(dst + 16 + klen, vals + val_off[i], vlen, lane);
+    }
+}
+
+extern "C" int dbeel_gpu_encode_run(uint64_t n_entries, const uint8_t* keys,
+                                    const uint64_t* key_offsets,
+                                    const uint8_t* values,
+                                    const uint64_t* value_offsets,
+                                    const uint8_t* timestamps, int device,
+                                    dbeel_compact_result* out) {
+    g_err[0] = 0;
+    if (!out || (n_entries && (!keys || !key_offsets || !values ||
+                               !value_offsets || !timestamps))) {
+        set_err("encode_run: null argument");
+        return DBEEL_ERR_INVALID_ARG;
+    }
+    memset(out, 0, sizeof *out);
+    if (device < 0) {
+        set_err("device must be >= 0 (no CPU fallback)");
+        return DBEEL_ERR_INVALID_ARG;
+    }
+    int ndev = 0;
+    hipError_t de = hipGetDeviceCount(&ndev);
+    if (de != hipSuccess || device >= ndev) {
+        set_err("no usable HIP device %d", device);
+        return DBEEL_ERR_NO_GPU;
+    }
+    HIP_CHECK(hipSetDevice(device));
+    if (n_entries == 0) {
+        out->data = (uint8_t*)malloc(1);
+        out->index = (uint8_t*)malloc(1);
+        return DBEEL_OK;
+    }
+    uint64_t n = n_entries;
+    uint64_t kbytes = key_offsets[n], vbytes = value_offsets[n];
+    for (uint64_t i = 0; i < n; i++) {
+        uint64_t klen = key_offsets[i + 1] - key_offsets[i];
+        uint64_t vlen = value_offsets[i + 1] - value_offsets[i];
+        if (key_offsets[i + 1] < key_offsets[i] ||
+            value_offsets[i + 1] < value_offsets[i] ||
+            32 + klen + vlen > 0xFFFFFFFFull) {
+            set_err("entry %llu too large or offsets not monotone",
+                    (unsigned long long)i);
+            return DBEEL_ERR_ITEM_TOO_LARGE; /* error.rs:60-61 analogue */
+        }
+    }
+
+    uint8_t *d_keys = nullptr, *d_vals = nullptr, *d_ts = nullptr,
+            *d_outd = nullptr, *d_outi = nullptr;
+    uint64_t *d_koff = nullptr, *d_voff = nullptr, *d_sizes = nullptr,
+             *d_doff = nullptr;
+    void* d_tmp = nullptr;
+    hipStream_t s = nullptr;
+    uint64_t total = 0;
+    int rc = DBEEL_OK;
+    size_t tmp_bytes = 0;
+
+#define ENC_CHECK(call)                                                     \
+    do {                                                                    \
+        hipError_t _e = (call);                                             \
+        if (_e != hipSuccess) {                                             \
+            set_err("%s failed: %s", #call, hipGetErrorString(_e));         \
+            rc = (_e == hipErrorOutOfMemory) ? DBEEL_ERR_OOM                \
+                                             : DBEEL_ERR_HIP;               \
+            goto done;                                                      \
+        }                                                                   \
+    } while (0)
+
+    ENC_CHECK(hipStreamCreate(&s));
+    ENC_CHECK(hipMalloc(&d_keys, kbytes ? kbytes : 1));
+    ENC_CHECK(hipMalloc(&d_vals, vbytes ? vbytes : 1));
+    ENC_CHECK(hipMalloc(&d_ts, n * 16));
+    ENC_CHECK(hipMalloc(&d_koff, (n + 1) * 8));
+    ENC_CHECK(hipMalloc(&d_voff, (n + 1) * 8));
+    ENC_CHECK(hipMalloc(&d_sizes, n * 8));
+    ENC_CHECK(hipMalloc(&d_doff, n * 8));
+    ENC_CHECK(hipMemcpyAsync(d_keys, keys, kbytes ? kbytes : 1,
+                             hipMemcpyHostToDevice, s));
+    ENC_CHECK(hipMemcpyAsync(d_vals, values, vbytes ? vbytes : 1,
+                             hipMemcpyHostToDevice, s));
+    ENC_CHECK(hipMemcpyAsync(d_ts, timestamps, n * 16,
+                             hipMemcpyHostToDevice, s));
+    ENC_CHECK(hipMemcpyAsync(d_koff, key_offsets, (n + 1) * 8,
+                             hipMemcpyHostToDevice, s));
+    ENC_CHECK(hipMemcpyAsync(d_voff, value_offsets, (n + 1) * 8,
+                             hipMemcpyHostToDevice, s));
+    hipLaunchKernelGGL(k_encode_sizes, dim3(pick_grid(n, 256)), dim3(256), 0,
+                       s, n, d_koff, d_voff, d_sizes);
+    (void)rocprim::exclusive_scan(nullptr, tmp_bytes, d_sizes, d_doff,
+                                  (uint64_t)0, n, rocprim::plus<uint64_t>(),
+                                  s);
+    ENC_CHECK(hipMalloc(&d_tmp, tmp_bytes));
+    (void)rocprim::exclusive_scan(d_tmp, tmp_bytes, d_sizes, d_doff,
+                                  (uint64_t)0, n, rocprim::plus<uint64_t>(),
+                                  s);
+    total = 32 * n + kbytes + vbytes; /* 32+klen+vlen per entry */
+    ENC_CHECK(hipMalloc(&d_outd, total));
+    ENC_CHECK(hipMalloc(&d_outi, n * 16));
+    hipLaunchKernelGGL(k_encode_index, dim3(pick_grid(n, 256)), dim3(256), 0,
+                       s, n, d_koff, d_doff, d_sizes, d_outi);
+    hipLaunchKernelGGL(k_encode_data, dim3(pick_grid(n * 64, 256)), dim3(256),
+                       0, s, n, d_keys, d_koff, d_vals, d_voff, d_ts, d_doff,
+                       d_outd);
+    ENC_CHECK(hipStreamSynchronize(s));
+    ENC_CHECK(hipGetLastError());
+
+    out->data = (uint8_t*)malloc(total ? total : 1);
+    out->index = (uint8_t*)malloc(n * 16);
+    if (!out->data || !out->index) {
+        free(out->data);
+        free(out->index);
+        memset(out, 0, sizeof *out);
+        rc = DBEEL_ERR_OOM;
+        goto done;
+    }
+    ENC_CHECK(hipMemcpyAsync(out->data, d_outd, total,
+                             hipMemcpyDeviceToHost, s));
+    ENC_CHECK(hipMemcpyAsync(out->index, d_outi, n * 16,
+                             hipMemcpyDeviceToHost, s));
+    ENC_CHECK(hipStreamSynchronize(s));
+    out->data_len = total;
+    out->index_len = n * 16;
+    out->entries_written = n;
+done:
+    hipFree(d_keys);
+    hipFree(d_vals);
+    hipFree(d_ts);
+    hipFree(d_koff);
+    hipFree(d_voff);
+    hipFree(d_sizes);
+    hipFree(d_doff);
+    hipFree(d_tmp);
+    hipFree(d_outd);
+    hipFree(d_outi);
+    if (s) hipStreamDestroy(s);
+    if (rc != DBEEL_OK && out->data) {
+        free(out->data);
+        free(out->index);
+        memset(out, 0, sizeof *out);
+    }
+    return rc;
+#undef ENC_CHECK
+}
+
+/* ------------------------------------------------------------------ */
 /* Host side                                                          */
 /* ------------------------------------------------------------------ */
 

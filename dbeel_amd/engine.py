@@ -103,6 +103,14 @@ def load() -> ctypes.CDLL:
             ctypes.c_void_p, ctypes.POINTER(CompactResult)
         ]
         lib.dbeel_gpu_job_destroy.argtypes = [ctypes.c_void_p]
+        lib.dbeel_gpu_encode_run.restype = ctypes.c_int
+        lib.dbeel_gpu_encode_run.argtypes = [
+            ctypes.c_uint64,
+            ctypes.POINTER(ctypes.c_uint8), ctypes.POINTER(ctypes.c_uint64),
+            ctypes.POINTER(ctypes.c_uint8), ctypes.POINTER(ctypes.c_uint64),
+            ctypes.POINTER(ctypes.c_uint8), ctypes.c_int,
+            ctypes.POINTER(CompactResult),
+        ]
         _lib = lib
     return _lib
 
@@ -152,6 +160,46 @@ def compact(runs, keep_tombstones: bool, device: int = 0,
     del keepalive
     if want_timings:
         return data, index, n, tim.as_dict()
+    return data, index, n
+
+
+def encode_run(entries, device: int = 0):
+    """GPU run encoder (the memtable-flush path, lsm_tree.rs:925-946):
+    entries = iterable of (key: bytes, data: bytes, timestamp: int),
+    ALREADY sorted ascending by key (memtable order). Returns
+    (data_bytes, index_bytes, n)."""
+    lib = load()
+    keys = b"".join(e[0] for e in entries)
+    vals = b"".join(e[1] for e in entries)
+    n = len(entries)
+    koff = np.zeros(n + 1, dtype=np.uint64)
+    voff = np.zeros(n + 1, dtype=np.uint64)
+    ts = np.zeros(16 * n, dtype=np.uint8)
+    for i, (k, v, t) in enumerate(entries):
+        koff[i + 1] = koff[i] + len(k)
+        voff[i + 1] = voff[i] + len(v)
+        ts[16 * i : 16 * (i + 1)] = np.frombuffer(
+            int(t).to_bytes(16, "little", signed=True), dtype=np.uint8
+        )
+    ka = np.frombuffer(keys or b"\0", dtype=np.uint8)
+    va = np.frombuffer(vals or b"\0", dtype=np.uint8)
+    res = CompactResult()
+    rc = lib.dbeel_gpu_encode_run(
+        n,
+        ka.ctypes.data_as(ctypes.POINTER(ctypes.c_uint8)),
+        koff.ctypes.data_as(ctypes.POINTER(ctypes.c_uint64)),
+        va.ctypes.data_as(ctypes.POINTER(ctypes.c_uint8)),
+        voff.ctypes.data_as(ctypes.POINTER(ctypes.c_uint64)),
+        ts.ctypes.data_as(ctypes.POINTER(ctypes.c_uint8)),
+        device, ctypes.byref(res),
+    )
+    if rc != 0:
+        raise DbeelGpuError(rc, lib.dbeel_gpu_last_error().decode())
+    try:
+        data = ctypes.string_at(res.data, res.data_len) if res.data_len else b""
+        index = ctypes.string_at(res.index, res.index_len) if res.index_len else b""
+    finally:
+        lib.dbeel_gpu_result_free(ctypes.byref(res))
     return data, index, n
 
 

@@ -113,6 +113,20 @@ void dbeel_gpu_result_free(dbeel_compact_result* r);
 /* Thread-local message for the last error in this thread. */
 const char* dbeel_gpu_last_error(void);
 
+/* ---- Run encoder (the memtable-flush path) ----
+ * Encodes an already-sorted (key, value, timestamp) stream into a run:
+ * the same entry layout + 16-byte index records flush_memtable_to_disk /
+ * EntryWriter produce (lsm_tree.rs:925-946, entry_writer.rs:71-98).
+ * keys/values: concatenated byte blobs with n+1 exclusive offsets;
+ * timestamps: n x 16-byte little-endian i128. Outputs are engine-allocated
+ * (dbeel_gpu_result_free). Oversized entries (full_size > u32::MAX) return
+ * DBEEL_ERR_ITEM_TOO_LARGE (error.rs:60-61). */
+int dbeel_gpu_encode_run(uint64_t n_entries, const uint8_t* keys,
+                         const uint64_t* key_offsets, const uint8_t* values,
+                         const uint64_t* value_offsets,
+                         const uint8_t* timestamps, int device,
+                         dbeel_compact_result* out);
+
 /* ---- Resident-job API (benchmarking / repeated compactions) ----
  * Uploads the runs to `device` once; each run executes the device pipeline
  * with inputs already resident in HBM (what BASELINE's MB/s is quoted on)

@@ -15,10 +15,18 @@ HEADER_SYMBOLS = [
     "dbeel_gpu_compact_timed",
     "dbeel_gpu_result_free",
     "dbeel_gpu_last_error",
+    "dbeel_gpu_encode_run",
     "dbeel_gpu_job_create",
     "dbeel_gpu_job_run",
     "dbeel_gpu_job_fetch",
     "dbeel_gpu_job_destroy",
+]
+
+LSM_SYMBOLS = [
+    "dbeel_lsm_compact",
+    "dbeel_lsm_replay",
+    "dbeel_lsm_compact_tree",
+    "dbeel_bloom_contains",
 ]
 
 
@@ -40,6 +48,14 @@ def test_header_declares_every_symbol():
     hdr = open(os.path.join(REPO_ROOT, "include", "dbeel_gpu.h")).read()
     for sym in HEADER_SYMBOLS:
         assert sym in hdr, sym
+
+
+def test_lsm_header_and_exports():
+    hdr = open(os.path.join(REPO_ROOT, "include", "dbeel_lsm.h")).read()
+    lib = ctypes.CDLL(_built())
+    for sym in LSM_SYMBOLS:
+        assert sym in hdr, sym
+        assert getattr(lib, sym, None) is not None, sym
 
 
 def test_device_minus_one_rejected():

@@ -140,6 +140,37 @@ def test_large_scale_properties(engine):
     assert not any(e.is_tombstone for e in ents)
 
 
+def test_encode_run_matches_entry_writer_layout(engine):
+    """GPU flush encoder: byte parity with the
+    EntryWriter layout (entry_writer.rs:71-98) as pinned by
+    dbeel_amd.format.build_run, incl. tombstones, empty keys, negative
+    timestamps and ragged sizes."""
+    from dbeel_amd.engine import encode_run
+    from dbeel_amd.format import Entry, build_run
+
+    rng = np.random.default_rng(31)
+    ents = []
+    keys = {b""} | {
+        bytes(rng.integers(0, 256, int(rng.integers(1, 100)), dtype=np.uint8))
+        for _ in range(3000)
+    }
+    for i, k in enumerate(sorted(keys)):
+        dlen = int(rng.integers(0, 5)) * 211
+        data = bytes(rng.integers(0, 256, dlen, dtype=np.uint8))
+        ents.append((k, data, int(rng.integers(-(10**18), 10**18))))
+
+    gd, gi, gn = encode_run(ents, device=0)
+    ed, ei = build_run([Entry(*e) for e in ents])
+    assert gn == len(ents)
+    assert gi == ei
+    assert gd == ed
+
+    # flush -> compact composition: an encoded run feeds compaction
+    data2, index2, n2 = engine.compact([(gd, gi)], keep_tombstones=True,
+                                       device=0)
+    assert (data2, index2) == (gd, gi)  # single sorted run is idempotent
+
+
 def test_resident_job_repeatable(engine):
     """Job API: repeated runs on resident inputs give identical results and
     both keep_tombstones settings work on one upload."""
