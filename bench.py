@@ -97,13 +97,24 @@ def main():
 
     jobs = [dbeel_amd.Job(runs, device=local_rank) for runs in job_runs]
 
+    pool = None
+    if len(jobs) > 1:
+        # independent jobs overlap on their own HIP streams (one dbeel
+        # shard each — SURVEY.md §8e); ctypes releases the GIL during the
+        # blocking engine calls
+        from concurrent.futures import ThreadPoolExecutor
+
+        pool = ThreadPoolExecutor(max_workers=len(jobs))
+
     def run_all():
-        tb = te = 0
+        if pool:
+            results = list(pool.map(lambda j: j.run(keep), jobs))
+        else:
+            results = [jobs[0].run(keep)]
+        tb = sum(r[0] for r in results)
+        te = sum(r[1] for r in results)
         t_acc = None
-        for j in jobs:
-            ob, oe, t = j.run(keep)
-            tb += ob
-            te += oe
+        for _, _, t in results:
             if t_acc is None:
                 t_acc = dict(t)
             else:
