@@ -223,18 +223,31 @@ def main():
         bs = args.cpu_baseline_scale * args.scale
         cjobs, _ = make_workload(0, bs, args.workload)
         cbytes = sum(d.nbytes + i.nbytes for runs in cjobs for d, i in runs)
+        # dbeel runs one single-threaded compaction per shard core
+        # (SURVEY.md §8d): 1 core for a single job, one core per job for
+        # the multi-job config
+        cores = min(len(cjobs), os.cpu_count() or 1)
         c0 = time.perf_counter()
-        for cruns in cjobs:
-            oracle.compact(cruns, keep_tombstones=keep)
+        if cores > 1:
+            from concurrent.futures import ThreadPoolExecutor
+
+            with ThreadPoolExecutor(max_workers=cores) as tp:
+                list(tp.map(
+                    lambda r: oracle.compact(r, keep_tombstones=keep), cjobs
+                ))
+        else:
+            for cruns in cjobs:
+                oracle.compact(cruns, keep_tombstones=keep)
         c1 = time.perf_counter()
         cpu_baseline = {
             "value": round((cbytes / 1e6) / (c1 - c0), 1),
             "unit": "MB/s",
-            "cores": 1,
+            "cores": cores,
             "kind": "port",
             "sample": (
                 f"{args.workload} shape at {bs:.2f} scale "
-                f"({cbytes / 1e6:.0f} MB input, {c1 - c0:.1f}s on 1 core; "
+                f"({cbytes / 1e6:.0f} MB input, {c1 - c0:.1f}s on "
+                f"{cores} core(s), one single-threaded job per core; "
                 "C oracle restatement — Rust/glommio unbuildable here, "
                 "BASELINE.md)"
             ),

@@ -5,22 +5,31 @@
  * restated in the header). This is a from-scratch GPU design, not a port of
  * the reference's heap loop:
  *
- *   k_validate : each run strictly sorted by key, entries well-formed
- *                (dbeel flush invariant, lsm_tree.rs:925-946; corrupt input
- *                errors loudly where the reference silently truncates)
- *   k_rank     : every entry's global merge rank by per-run binary search
- *                (order = key bytes asc, timestamp i128 asc, run index asc —
- *                lsm_tree.rs:52-71 + mod.rs:75-81); the winner test
- *                ("no other run holds an equal key later in the order",
- *                lsm_tree.rs:1041-1044) falls out of the same searches free.
- *   scans      : rocPRIM exclusive scans of survivor sizes/flags ->
- *                output byte offsets + positions.
- *   k_emit     : output .index records (offset/key_size/full_size,
- *                entry_writer.rs:79-87) + compacted source map.
- *   k_copy     : verbatim survivor byte copy, balanced by DESTINATION
- *                granule (one 16-B granule per lane, 4-KiB window per
- *                256-thread block) so throughput is independent of entry
- *                size; aligned 16-B stores, unaligned 16-B loads.
+ *   k_prepare    : validates every entry (bounds + bincode field
+ *                  cross-check; corrupt input errors loudly where the
+ *                  reference silently truncates) and extracts the dense
+ *                  key-prefix (8 B) and aux (64 B: klen/ts/first-40-key-
+ *                  bytes) arrays the merge runs on.
+ *   k_corank     : block-cooperative merge-path co-ranking — for every
+ *                  run pair, diagonal-partitioned 4096-position windows;
+ *                  blocks stage both prefix segments into LDS coalesced
+ *                  and walk 16 merged positions per thread, recording
+ *                  crossranks (order = key bytes asc, timestamp i128 asc,
+ *                  run index asc — lsm_tree.rs:52-71 + mod.rs:75-81) and
+ *                  newest-wins supersession flags (lsm_tree.rs:1041-1044).
+ *   k_rankreduce : crossranks -> global rank; winner/tombstone rules;
+ *                  strict-sortedness check (flush invariant,
+ *                  lsm_tree.rs:925-946); one 32-B rank-indexed record.
+ *   scans        : rocPRIM exclusive scans (transform iterators) ->
+ *                  survivor byte offsets + positions.
+ *   k_emit       : output .index records (offset/key_size/full_size,
+ *                  entry_writer.rs:79-87) + compacted source map.
+ *   k_winmap/k_copy : verbatim survivor copy, balanced by DESTINATION
+ *                  granule (4 x 16-B granules per lane, 16-KiB window per
+ *                  256-thread block, LDS granule->entry map) so throughput
+ *                  is independent of entry size; aligned non-temporal
+ *                  16-B stores, unaligned 16-B loads.
+ *   k_encode_*   : the memtable-flush run encoder (lsm_tree.rs:925-946).
  *
  * All integer/byte work, HBM-bandwidth bound; MFMA unused by design
  * (BASELINE.json north_star).
@@ -646,7 +655,7 @@ __global__ __launch_bounds__(COPY_BLOCK) void k_copy(
         __syncthreads();
 
         #pragma unroll
-        for (int q = 0; q < 4; q++) {
+        for (int q = 0; q < COPY_GRANULES / COPY_BLOCK; q++) {
             uint32_t gl = q * COPY_BLOCK + threadIdx.x; /* window-local */
             uint64_t gpos = wstart + (uint64_t)gl * 16;
             if (gpos >= total_bytes) break;
