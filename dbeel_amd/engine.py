@@ -66,6 +66,15 @@ class CompactTimings(ctypes.Structure):
         return {k: getattr(self, k) for k, _ in self._fields_}
 
 
+def _ptr_bytes(ptr, n: int) -> bytes:
+    """Copy n bytes from a ctypes uint8 pointer. ctypes.string_at truncates
+    its size argument to 32 bits on this interpreter (CPython 3.10), which
+    silently corrupts >4 GB results — use a numpy view instead."""
+    if not n:
+        return b""
+    return np.ctypeslib.as_array(ptr, shape=(int(n),)).tobytes()
+
+
 _lib = None
 
 
@@ -152,8 +161,8 @@ def compact(runs, keep_tombstones: bool, device: int = 0,
     if rc != 0:
         raise DbeelGpuError(rc, lib.dbeel_gpu_last_error().decode())
     try:
-        data = ctypes.string_at(res.data, res.data_len) if res.data_len else b""
-        index = ctypes.string_at(res.index, res.index_len) if res.index_len else b""
+        data = _ptr_bytes(res.data, res.data_len)
+        index = _ptr_bytes(res.index, res.index_len)
         n = int(res.entries_written)
     finally:
         lib.dbeel_gpu_result_free(ctypes.byref(res))
@@ -196,8 +205,8 @@ def encode_run(entries, device: int = 0):
     if rc != 0:
         raise DbeelGpuError(rc, lib.dbeel_gpu_last_error().decode())
     try:
-        data = ctypes.string_at(res.data, res.data_len) if res.data_len else b""
-        index = ctypes.string_at(res.index, res.index_len) if res.index_len else b""
+        data = _ptr_bytes(res.data, res.data_len)
+        index = _ptr_bytes(res.index, res.index_len)
     finally:
         lib.dbeel_gpu_result_free(ctypes.byref(res))
     return data, index, n
@@ -237,8 +246,8 @@ class Job:
         if rc != 0:
             raise DbeelGpuError(rc, self._lib.dbeel_gpu_last_error().decode())
         try:
-            data = ctypes.string_at(res.data, res.data_len) if res.data_len else b""
-            index = ctypes.string_at(res.index, res.index_len) if res.index_len else b""
+            data = _ptr_bytes(res.data, res.data_len)
+            index = _ptr_bytes(res.index, res.index_len)
             n = int(res.entries_written)
         finally:
             self._lib.dbeel_gpu_result_free(ctypes.byref(res))

@@ -40,6 +40,15 @@ def build(force: bool = False) -> str:
     return _LIB
 
 
+def _ptr_bytes(ptr, n: int) -> bytes:
+    """Copy n bytes from a ctypes uint8 pointer. ctypes.string_at truncates
+    its size argument to 32 bits on this interpreter (CPython 3.10), which
+    silently corrupts >4 GB results — use a numpy view instead."""
+    if not n:
+        return b""
+    return np.ctypeslib.as_array(ptr, shape=(int(n),)).tobytes()
+
+
 _lib = None
 
 
@@ -93,8 +102,8 @@ def compact(runs, keep_tombstones: bool) -> tuple[bytes, bytes, int]:
             f"oracle error {rc}: {lib.dbeel_oracle_last_error().decode()}"
         )
     try:
-        data = ctypes.string_at(res.data, res.data_len) if res.data_len else b""
-        index = ctypes.string_at(res.index, res.index_len) if res.index_len else b""
+        data = _ptr_bytes(res.data, res.data_len)
+        index = _ptr_bytes(res.index, res.index_len)
         n = int(res.entries_written)
     finally:
         lib.dbeel_oracle_result_free(ctypes.byref(res))

@@ -123,10 +123,53 @@ def test_corrupt_input_rejected(engine):
     assert ei.value.code == 2
 
 
+def test_full_cfg3_exact_parity(engine):
+    """FULL BASELINE config 3 (8 x 1 GiB runs, 50% overlap, 5% tombstones):
+    exact byte parity of the 6.1 GB output vs the CPU oracle, plus the
+    size-independent properties (sortedness enforced by feeding the output
+    back through the engine, which validates the flush invariant)."""
+    from dbeel_amd.genruns import make_config
+
+    runs = make_config("cfg3")
+    gd, gi, gn = engine.compact(runs, keep_tombstones=False, device=0)
+    od, oi, on = oracle.compact(runs, keep_tombstones=False)
+    assert gn == on
+    assert gi == oi
+    assert gd == od
+    # output is itself a valid sorted unique-key run: recompacting the
+    # single run must be byte-idempotent (k_rankreduce would error on any
+    # sortedness violation)
+    rd, ri, rn = engine.compact([(gd, gi)], keep_tombstones=True, device=0)
+    assert (rd, ri, rn) == (gd, gi, gn)
+
+
+def test_sixtyfour_runs_bound(engine):
+    """MAX_RUNS=64 merge (the ABI bound); 65 runs must be rejected."""
+    from dbeel_amd.engine import DbeelGpuError
+    from dbeel_amd.format import Entry, build_run
+
+    rng = np.random.default_rng(77)
+    runs = []
+    for r in range(64):
+        keys = sorted({bytes(rng.integers(0, 256, 6, dtype=np.uint8))
+                       for _ in range(30)})
+        runs.append(build_run(
+            [Entry(k, bytes([r]), (r << 20) + i)
+             for i, k in enumerate(keys)]
+        ))
+    gd, gi, gn = engine.compact(runs, keep_tombstones=True, device=0)
+    od, oi, on = oracle.compact(runs, keep_tombstones=True)
+    assert (gd, gi, gn) == (od, oi, on)
+
+    with pytest.raises(DbeelGpuError) as ei:
+        engine.compact(runs + [runs[0]], keep_tombstones=True, device=0)
+    assert ei.value.code == 1
+
+
 def test_large_scale_properties(engine):
-    """Mid-scale run (~600 MB input) — parity vs oracle would be slow at
-    full BASELINE size, so check parity at this size and properties that are
-    size-independent: sortedness, unique keys, verbatim bytes, count."""
+    """Mid-scale run (~600 MB input) — byte parity vs oracle plus
+    size-independent properties: sortedness, unique keys, verbatim bytes,
+    count."""
     runs = make_runs(8, 120_000, 32, 1024, overlap_frac=0.5,
                      tombstone_frac=0.05, seed=0xC0FFEE)
     gd, gi, gn = engine.compact(runs, keep_tombstones=False, device=0)
