@@ -130,40 +130,6 @@ uint64_t fnv1a64(const uint8_t* p, size_t n, uint64_t seed) {
     return h;
 }
 
-void bloom_build(const std::vector<uint8_t>& out_data,
-                 const std::vector<uint8_t>& out_index, uint64_t n_items,
-                 std::vector<uint8_t>& file) {
-    /* sizing for 1% fp (BLOOM_MAX_ALLOWED_ERROR, lsm_tree.rs:48):
-     * n_bits = n * ln(0.01)/-ln(2)^2 ~= 9.585 n; k = 7 */
-    if (n_items == 0) n_items = 1;
-    uint64_t n_bits = (uint64_t)(9.585 * (double)n_items) + 64;
-    uint64_t n_bytes = (n_bits + 7) / 8;
-    BloomHeader h{};
-    memcpy(h.magic, "DBLM", 4);
-    h.version = 1;
-    h.k = 7;
-    h.n_bits = n_bits;
-    h.seed = 0xDBEE1;
-    file.resize(sizeof h + n_bytes, 0);
-    memcpy(file.data(), &h, sizeof h);
-    uint8_t* bits = file.data() + sizeof h;
-    size_t n = out_index.size() / 16;
-    for (size_t i = 0; i < n; i++) {
-        uint64_t off, klen;
-        uint32_t key_size;
-        memcpy(&off, out_index.data() + i * 16, 8);
-        memcpy(&key_size, out_index.data() + i * 16 + 8, 4);
-        klen = key_size - 8;
-        const uint8_t* key = out_data.data() + off + 8;
-        uint64_t h1 = fnv1a64(key, klen, h.seed);
-        uint64_t h2 = fnv1a64(key, klen, h.seed ^ 0x9e3779b97f4a7c15ull) | 1;
-        for (uint32_t j = 0; j < h.k; j++) {
-            uint64_t bit = (h1 + j * h2) % n_bits;
-            bits[bit / 8] |= (uint8_t)(1u << (bit % 8));
-        }
-    }
-}
-
 } // namespace
 
 extern "C" int dbeel_bloom_contains(const uint8_t* bloom_bytes,
@@ -263,29 +229,37 @@ extern "C" int dbeel_lsm_compact(const char* dir_c, const uint64_t* indices,
         total_input_data += datas[i].size();
     }
 
-    dbeel_compact_result res{};
-    int rc = dbeel_gpu_compact(views.data(), n_indices, keep_tombstones,
-                               device, &res);
+    dbeel_gpu_job* job = nullptr;
+    int rc = dbeel_gpu_job_create(views.data(), n_indices, device, &job);
     if (rc != DBEEL_OK) return rc;
+    rc = dbeel_gpu_job_run(job, keep_tombstones, nullptr, nullptr, nullptr);
+    dbeel_compact_result res{};
+    if (rc == DBEEL_OK) rc = dbeel_gpu_job_fetch(job, &res);
+
+    /* bloom only when input data exceeds the threshold
+     * (lsm_tree.rs:1026-1034; default 1 MiB, mod.rs:19) — built on the
+     * device from the job's resident output */
+    bool with_bloom = total_input_data > sstable_bloom_min_size;
+    uint8_t* bloom = nullptr;
+    uint64_t bloom_len = 0;
+    if (rc == DBEEL_OK && with_bloom)
+        rc = dbeel_gpu_job_bloom(job, &bloom, &bloom_len);
+    dbeel_gpu_job_destroy(job);
+    if (rc != DBEEL_OK) {
+        dbeel_gpu_result_free(&res);
+        dbeel_gpu_bloom_free(bloom);
+        return rc;
+    }
 
     std::string cd = file_path(dir, output_index, "compact_data");
     std::string ci = file_path(dir, output_index, "compact_index");
     std::string cb = file_path(dir, output_index, "compact_bloom");
     bool ok = write_whole(cd, res.data, res.data_len) &&
               write_whole(ci, res.index, res.index_len);
-
-    /* bloom only when input data exceeds the threshold
-     * (lsm_tree.rs:1026-1034; default 1 MiB, mod.rs:19) */
-    bool with_bloom = total_input_data > sstable_bloom_min_size;
-    if (ok && with_bloom) {
-        std::vector<uint8_t> od(res.data, res.data + res.data_len);
-        std::vector<uint8_t> oi(res.index, res.index + res.index_len);
-        std::vector<uint8_t> bloom;
-        bloom_build(od, oi, res.entries_written, bloom);
-        ok = write_whole(cb, bloom.data(), bloom.size());
-    }
+    if (ok && with_bloom) ok = write_whole(cb, bloom, bloom_len);
     uint64_t written = res.entries_written;
     dbeel_gpu_result_free(&res);
+    dbeel_gpu_bloom_free(bloom);
     if (!ok) return DBEEL_ERR_HIP;
 
     /* journal (CompactionAction, lsm_tree.rs:1090-1105): renames of all
@@ -417,12 +391,18 @@ extern "C" int dbeel_lsm_compact_tree(const char* dir_c,
             it2->second.insert(it2->second.end(), items.begin(),
                                items.end());
     }
-    /* deterministic final order: biggest size class first ->
-     * keep_tombstones = (i > 0) drops tombstones only on the final level
-     * (tasks/compaction.rs:92) */
+    /* deterministic final order: biggest size class first */
     std::sort(optimized.begin(), optimized.end(),
               [](auto& a, auto& b) { return a.first < b.first; });
 
+    /* Tombstone rule: the reference drops tombstones "only on the final
+     * level" via keep_tombstones = (i > 0) over a HashMap enumeration
+     * (tasks/compaction.rs:82-92) — nondeterministic, and a promoted
+     * group of NEW runs can enumerate first and drop tombstones while
+     * older runs still hold the deleted keys (data resurrection). This
+     * port implements the comment's intent deterministically: drop
+     * tombstones only when the group covers EVERY live sstable (nothing
+     * outside it can resurrect). Documented divergence. */
     uint32_t done = 0;
     for (size_t i = 0; i < optimized.size(); i++) {
         auto& items = optimized[i].second;
@@ -430,8 +410,9 @@ extern "C" int dbeel_lsm_compact_tree(const char* dir_c,
         std::vector<uint64_t> idxs = items;
         std::sort(idxs.begin(), idxs.end()); /* ascending sstable index —
                                                 the tie-break needs it */
+        int keep = (items.size() == tables.size()) ? 0 : 1;
         int rc = dbeel_lsm_compact(dir_c, idxs.data(), idxs.size(),
-                                   out_index, i > 0 ? 1 : 0, device,
+                                   out_index, keep, device,
                                    sstable_bloom_min_size, nullptr);
         if (rc != DBEEL_OK) return rc;
         out_index += 2;

@@ -70,12 +70,12 @@ typedef struct {
  * (rank -> scan -> emit -> copy), excluding host<->device copies. */
 typedef struct {
     double h2d_ms;
-    double prep_ms;    /* k_prepare: validation + key-prefix extract,
-                          k_pivots: crossrank pivots                 */
-    double rank_ms;    /* k_rank: global merge-rank + winner flags  */
-    double scan_ms;    /* size/position prefix sums                 */
+    double prep_ms;    /* k_prepare: validation + key-prefix/aux extract */
+    double rank_ms;    /* k_corank + k_rankreduce: merge-path crossranks,
+                          global ranks, winner flags                 */
+    double scan_ms;    /* survivor size/position prefix sums        */
     double emit_ms;    /* output index build + survivor source map  */
-    double copy_ms;    /* verbatim entry copy-out                   */
+    double copy_ms;    /* verbatim entry copy-out (incl. k_winmap)  */
     double kernel_ms;  /* prep+rank+scan+emit+copy                  */
     double d2h_ms;
 } dbeel_compact_timings;
@@ -141,6 +141,13 @@ int dbeel_gpu_job_run(dbeel_gpu_job* job, int keep_tombstones,
                       dbeel_compact_timings* t);
 /* Copies the last run's outputs to host (engine-allocated). */
 int dbeel_gpu_job_fetch(dbeel_gpu_job* job, dbeel_compact_result* out);
+/* Builds the behavioral "DBLM" bloom over the last run's surviving keys
+ * on the device (the Bloom::set-per-written-key step,
+ * lsm_tree.rs:1049-1051; format in include/dbeel_lsm.h). Engine-allocated;
+ * free with dbeel_gpu_bloom_free. */
+int dbeel_gpu_job_bloom(dbeel_gpu_job* job, uint8_t** out_bytes,
+                        uint64_t* out_len);
+void dbeel_gpu_bloom_free(uint8_t* p);
 void dbeel_gpu_job_destroy(dbeel_gpu_job* job);
 
 #ifdef __cplusplus

@@ -50,11 +50,12 @@ int dbeel_lsm_replay(const char* dir, uint32_t* out_replayed);
  * sstables ({i:020}.index), group by leading_zeros(entry_count), promote
  * groups whose combined count reaches a bigger size class, compact every
  * group with >= max(2, compaction_factor) members into the next odd
- * output index (+2 per group). keep_tombstones = false only for the
- * LARGEST size class (the reference enumerates a HashMap here, which is
- * nondeterministic — tasks/compaction.rs:82-92; this port fixes the order
- * to descending size so "drop tombstones only on the final level" is
- * deterministic). */
+ * output index (+2 per group). Tombstones are dropped only when a group
+ * covers EVERY live sstable: the reference's keep_tombstones = (i > 0)
+ * over a HashMap enumeration (tasks/compaction.rs:82-92) is
+ * nondeterministic and can resurrect deleted keys when a promoted group
+ * of new runs enumerates first; this port implements the "only on the
+ * final level" intent deterministically and safely. */
 int dbeel_lsm_compact_tree(const char* dir, uint64_t compaction_factor,
                            int device, uint64_t sstable_bloom_min_size,
                            uint32_t* out_n_compactions);
