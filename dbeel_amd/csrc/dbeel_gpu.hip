@@ -1419,6 +1419,134 @@ extern "C" int dbeel_gpu_job_fetch(dbeel_gpu_job* job,
     return DBEEL_OK;
 }
 
+/* ------------------------------------------------------------------ */
+/* Batched point lookup                                               */
+/*                                                                    */
+/* GPU analogue of LSMTree::get over the sstables                     */
+/* (lsm_tree.rs:605-723, binary search per run lsm_tree.rs:674-723):  */
+/* one thread per query key binary-searches every run and keeps the   */
+/* newest match — max (timestamp, run index), the same winner rule as */
+/* compaction. Tombstones are reported as found with value_len 0 so   */
+/* the caller distinguishes deleted from absent (tests/db_server.rs   */
+/* delete->get->KeyNotFound semantics). Bloom prefiltering stays on   */
+/* the host (dbeel_bloom_contains).                                   */
+/* ------------------------------------------------------------------ */
+
+__global__ void k_lookup(RunsDesc R, const uint8_t* keys,
+                         const uint64_t* key_off, uint64_t n_keys,
+                         dbeel_lookup_hit* out) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t q = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         q < n_keys; q += stride) {
+        const uint8_t* key = keys + key_off[q];
+        uint64_t klen = key_off[q + 1] - key_off[q];
+        int best_run = -1;
+        uint64_t best_off = 0, best_vlen = 0;
+        uint64_t best_tslo = 0;
+        int64_t best_tshi = 0;
+        for (int r = 0; r < R.n_runs; r++) {
+            uint64_t lo = 0, hi = R.count[r];
+            while (lo < hi) {
+                uint64_t mid = (lo + hi) >> 1;
+                EView m;
+                if (!load_entry(R, r, mid, m)) {
+                    hi = mid; /* corrupt input; search degrades safely */
+                    continue;
+                }
+                int c = cmp_keys(m.key, m.klen, key, klen);
+                if (c < 0)
+                    lo = mid + 1;
+                else
+                    hi = mid;
+            }
+            if (lo >= R.count[r]) continue;
+            EView m;
+            if (!load_entry(R, r, lo, m)) continue;
+            if (m.klen != klen || cmp_keys(m.key, m.klen, key, klen) != 0)
+                continue;
+            uint64_t tslo;
+            int64_t tshi;
+            load_ts(m, tslo, tshi);
+            /* newest wins: (timestamp, run index); runs scan ascending so
+             * a later run with equal ts wins automatically */
+            if (best_run < 0 || tshi > best_tshi ||
+                (tshi == best_tshi && tslo >= best_tslo)) {
+                best_run = r;
+                best_off = m.off + 16 + m.klen; /* value bytes start */
+                best_vlen = (uint64_t)m.full_size - 32 - m.klen;
+                best_tslo = tslo;
+                best_tshi = tshi;
+            }
+        }
+        out[q].run = best_run;
+        out[q].is_tombstone = (best_run >= 0 && best_vlen == 0) ? 1 : 0;
+        out[q].value_offset = best_off;
+        out[q].value_len = best_vlen;
+    }
+}
+
+extern "C" int dbeel_gpu_lookup(const dbeel_run_view* runs, size_t n_runs,
+                                const uint8_t* keys,
+                                const uint64_t* key_offsets, uint64_t n_keys,
+                                int device, dbeel_lookup_hit* out) {
+    g_err[0] = 0;
+    if (!out || (n_keys && (!keys || !key_offsets))) {
+        set_err("lookup: null argument");
+        return DBEEL_ERR_INVALID_ARG;
+    }
+    int rc = validate_runs(runs, n_runs);
+    if (rc) return rc;
+    if (device < 0) {
+        set_err("device must be >= 0 (no CPU fallback)");
+        return DBEEL_ERR_INVALID_ARG;
+    }
+    int ndev = 0;
+    hipError_t de = hipGetDeviceCount(&ndev);
+    if (de != hipSuccess || device >= ndev) {
+        set_err("no usable HIP device %d", device);
+        return DBEEL_ERR_NO_GPU;
+    }
+    HIP_CHECK(hipSetDevice(device));
+    if (n_keys == 0) return DBEEL_OK;
+
+    dbeel_gpu_job* job = nullptr;
+    rc = dbeel_gpu_job_create(runs, n_runs, device, &job);
+    if (rc) return rc;
+
+    uint64_t kbytes = key_offsets[n_keys];
+    uint8_t* d_keys = nullptr;
+    uint64_t* d_koff = nullptr;
+    dbeel_lookup_hit* d_out = nullptr;
+    hipError_t e = hipMalloc(&d_keys, kbytes ? kbytes : 1);
+    if (e == hipSuccess) e = hipMalloc(&d_koff, (n_keys + 1) * 8);
+    if (e == hipSuccess)
+        e = hipMalloc(&d_out, n_keys * sizeof(dbeel_lookup_hit));
+    if (e == hipSuccess && kbytes)
+        e = hipMemcpyAsync(d_keys, keys, kbytes, hipMemcpyHostToDevice,
+                           job->stream);
+    if (e == hipSuccess)
+        e = hipMemcpyAsync(d_koff, key_offsets, (n_keys + 1) * 8,
+                           hipMemcpyHostToDevice, job->stream);
+    if (e == hipSuccess) {
+        hipLaunchKernelGGL(k_lookup, dim3(pick_grid(n_keys, 256)), dim3(256),
+                           0, job->stream, job->desc, d_keys, d_koff, n_keys,
+                           d_out);
+        e = hipMemcpyAsync(out, d_out, n_keys * sizeof(dbeel_lookup_hit),
+                           hipMemcpyDeviceToHost, job->stream);
+    }
+    if (e == hipSuccess) e = hipStreamSynchronize(job->stream);
+    if (e == hipSuccess) e = hipGetLastError();
+    hipFree(d_keys);
+    hipFree(d_koff);
+    hipFree(d_out);
+    dbeel_gpu_job_destroy(job);
+    if (e != hipSuccess) {
+        set_err("lookup failed: %s", hipGetErrorString(e));
+        return DBEEL_ERR_HIP;
+    }
+    return DBEEL_OK;
+}
+
 extern "C" int dbeel_gpu_compact_timed(const dbeel_run_view* runs,
                                        size_t n_runs, int keep_tombstones,
                                        int device, dbeel_compact_result* out,

@@ -172,6 +172,58 @@ def compact(runs, keep_tombstones: bool, device: int = 0,
     return data, index, n
 
 
+class LookupHit(ctypes.Structure):
+    _fields_ = [
+        ("run", ctypes.c_int32),
+        ("is_tombstone", ctypes.c_uint32),
+        ("value_offset", ctypes.c_uint64),
+        ("value_len", ctypes.c_uint64),
+    ]
+
+
+def lookup(runs, keys, device: int = 0):
+    """Batched point lookup (LSMTree::get over sstables,
+    lsm_tree.rs:605-723): for each key returns the value bytes of the
+    newest match, b"" for a deleted key (tombstone), or None if absent."""
+    lib = load()
+    if not hasattr(lib, "_lookup_ready"):
+        lib.dbeel_gpu_lookup.restype = ctypes.c_int
+        lib.dbeel_gpu_lookup.argtypes = [
+            ctypes.POINTER(RunView), ctypes.c_size_t,
+            ctypes.POINTER(ctypes.c_uint8), ctypes.POINTER(ctypes.c_uint64),
+            ctypes.c_uint64, ctypes.c_int, ctypes.POINTER(LookupHit),
+        ]
+        lib._lookup_ready = True
+    views, keepalive = _views(runs)
+    blob = b"".join(keys)
+    koff = np.zeros(len(keys) + 1, dtype=np.uint64)
+    np.cumsum([len(k) for k in keys], out=koff[1:])
+    ka = np.frombuffer(blob or b"\0", dtype=np.uint8)
+    hits = (LookupHit * len(keys))()
+    rc = lib.dbeel_gpu_lookup(
+        views, len(runs),
+        ka.ctypes.data_as(ctypes.POINTER(ctypes.c_uint8)),
+        koff.ctypes.data_as(ctypes.POINTER(ctypes.c_uint64)),
+        len(keys), device, hits,
+    )
+    if rc != 0:
+        raise DbeelGpuError(rc, lib.dbeel_gpu_last_error().decode())
+    out = []
+    datas = [bytes(_as_u8(d)) if not isinstance(d, bytes) else d
+             for d, _ in runs]
+    for h in hits:
+        if h.run < 0:
+            out.append(None)
+        elif h.is_tombstone:
+            out.append(b"")
+        else:
+            out.append(
+                datas[h.run][h.value_offset : h.value_offset + h.value_len]
+            )
+    del keepalive
+    return out
+
+
 def encode_run(entries, device: int = 0):
     """GPU run encoder (the memtable-flush path, lsm_tree.rs:925-946):
     entries = iterable of (key: bytes, data: bytes, timestamp: int),

@@ -113,6 +113,25 @@ void dbeel_gpu_result_free(dbeel_compact_result* r);
 /* Thread-local message for the last error in this thread. */
 const char* dbeel_gpu_last_error(void);
 
+/* ---- Batched point lookup ----
+ * GPU analogue of LSMTree::get over the sstables (lsm_tree.rs:605-723):
+ * for each query key, the newest match across the runs — max (timestamp,
+ * run index), the compaction winner rule. run = -1 when absent;
+ * is_tombstone = 1 distinguishes deleted from absent (the reference's
+ * delete->get->KeyNotFound semantics). value_offset indexes into
+ * runs[run].data. Bloom prefiltering stays host-side
+ * (dbeel_bloom_contains). keys: concatenated blob + n+1 offsets. */
+typedef struct {
+    int32_t run;
+    uint32_t is_tombstone;
+    uint64_t value_offset;
+    uint64_t value_len;
+} dbeel_lookup_hit;
+
+int dbeel_gpu_lookup(const dbeel_run_view* runs, size_t n_runs,
+                     const uint8_t* keys, const uint64_t* key_offsets,
+                     uint64_t n_keys, int device, dbeel_lookup_hit* out);
+
 /* ---- Run encoder (the memtable-flush path) ----
  * Encodes an already-sorted (key, value, timestamp) stream into a run:
  * the same entry layout + 16-byte index records flush_memtable_to_disk /

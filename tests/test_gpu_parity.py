@@ -214,6 +214,43 @@ def test_encode_run_matches_entry_writer_layout(engine):
     assert (data2, index2) == (gd, gi)  # single sorted run is idempotent
 
 
+def test_batched_lookup(engine):
+    """GPU-batched point lookup vs a host-side
+    newest-wins model: present, overwritten, deleted and absent keys,
+    plus the delete->get->KeyNotFound distinction
+    (tests/db_server.rs:183-213 semantics)."""
+    from dbeel_amd.engine import lookup
+    from dbeel_amd.format import parse_run
+
+    runs = make_runs(4, 3000, 16, 64, overlap_frac=0.5, tombstone_frac=0.15,
+                     seed=55)
+    runs = [(bytes(d), bytes(i)) for d, i in runs]
+
+    # host model: newest (ts, run) wins
+    model = {}
+    for r, (d, i) in enumerate(runs):
+        for e in parse_run(d, i):
+            cur = model.get(e.key)
+            if cur is None or (e.timestamp, r) > cur[0]:
+                model[e.key] = ((e.timestamp, r), e.data)
+
+    rng = np.random.default_rng(3)
+    present = list(model.keys())
+    queries = [present[int(rng.integers(0, len(present)))]
+               for _ in range(500)]
+    absent = [bytes(rng.integers(0, 256, 16, dtype=np.uint8))
+              for _ in range(100)]
+    queries += absent
+
+    got = lookup(runs, queries, device=0)
+    for k, g in zip(queries, got):
+        if k in model:
+            exp = model[k][1]  # b"" for tombstone-winner (deleted)
+            assert g == exp, k.hex()
+        else:
+            assert g is None, k.hex()
+
+
 def test_resident_job_repeatable(engine):
     """Job API: repeated runs on resident inputs give identical results and
     both keep_tombstones settings work on one upload."""
