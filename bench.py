@@ -78,13 +78,18 @@ def main():
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
     n_gpus = max(world, args.gpus)
+    # one rank per GPU; tolerate rehearsal runs with more ranks than GPUs
+    n_dev = torch.cuda.device_count() if torch.cuda.is_available() else 0
+    device = local_rank % n_dev if n_dev else local_rank
+    backend = os.environ.get("DBEEL_BENCH_BACKEND", "nccl")
     dist = None
     if world > 1:
         import torch.distributed as dist_mod
 
         dist = dist_mod
-        dist.init_process_group(backend="nccl")
-        torch.cuda.set_device(local_rank)
+        dist.init_process_group(backend=backend)
+        if n_dev:
+            torch.cuda.set_device(device)
 
     import dbeel_amd
 
@@ -95,7 +100,7 @@ def main():
     )
     n_entries = sum(i.nbytes // 16 for runs in job_runs for d, i in runs)
 
-    jobs = [dbeel_amd.Job(runs, device=local_rank) for runs in job_runs]
+    jobs = [dbeel_amd.Job(runs, device=device) for runs in job_runs]
 
     pool = None
     if len(jobs) > 1:
@@ -128,11 +133,12 @@ def main():
         out_bytes, out_entries, _ = run_all()
 
     def barrier_sync():
-        torch.cuda.synchronize(local_rank) if torch.cuda.is_available() else None
+        if torch.cuda.is_available():
+            torch.cuda.synchronize(device)
         if dist:
             dist.barrier()
         if torch.cuda.is_available():
-            torch.cuda.synchronize(local_rank)
+            torch.cuda.synchronize(device)
 
     barrier_sync()
     t0 = time.perf_counter()
@@ -143,8 +149,8 @@ def main():
         if dist:
             # the path's only collective: RCCL all-gather of emitted byte
             # counts over xGMI (north_star / SURVEY.md §5)
-            counts = torch.tensor([out_bytes], dtype=torch.int64,
-                                  device=f"cuda:{local_rank}")
+            dev = f"cuda:{device}" if backend == "nccl" else "cpu"
+            counts = torch.tensor([out_bytes], dtype=torch.int64, device=dev)
             gathered = [torch.zeros_like(counts) for _ in range(world)]
             dist.all_gather(gathered, counts)
     barrier_sync()
@@ -152,8 +158,8 @@ def main():
 
     # MAX over ranks
     if dist:
-        e = torch.tensor([elapsed], dtype=torch.float64,
-                         device=f"cuda:{local_rank}")
+        dev = f"cuda:{device}" if backend == "nccl" else "cpu"
+        e = torch.tensor([elapsed], dtype=torch.float64, device=dev)
         dist.all_reduce(e, op=dist.ReduceOp.MAX)
         elapsed = float(e.item())
 
