@@ -6,7 +6,7 @@ from __future__ import annotations
 import ctypes
 import os
 
-from .engine import ERROR_NAMES, DbeelGpuError, load as _load_gpu
+from .engine import DbeelGpuError, load as _load_gpu
 
 _lib = None
 
