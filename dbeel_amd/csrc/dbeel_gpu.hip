@@ -358,6 +358,11 @@ __global__ __launch_bounds__(CORANK_BLOCK) void k_corank(
     RunsDesc R, const uint64_t* pfx, const Aux* aux, const PairDesc* pairs,
     uint32_t n_pairs, uint64_t total_chunks, uint32_t* cr) {
     __shared__ uint64_t s_pfx[CORANK_BLOCK_POS + 2];
+    __shared__ uint32_t s_cr[CORANK_BLOCK_POS]; /* staged crossranks:
+        [0..lenA) for run a, [lenA..lenA+lenB) for run b — each block's
+        output ranges are contiguous, so results are staged here and
+        written back coalesced (direct interleaved 4-B stores measured
+        3x write amplification from line RMW thrash) */
     __shared__ uint64_t s_bounds[4]; /* iaS, ibS, iaE, ibE */
 
     for (uint64_t t = blockIdx.x; t < total_chunks; t += gridDim.x) {
@@ -435,11 +440,6 @@ __global__ __launch_bounds__(CORANK_BLOCK) void k_corank(
             uint32_t dend = d + CORANK_STEPS;
             if (dend > L) dend = L;
 
-            uint32_t sa = (uint32_t)(b < a ? b : b - 1); /* slot of b in a */
-            uint32_t sb = (uint32_t)(a < b ? a : a - 1); /* slot of a in b */
-            uint32_t* cra = cr + (uint64_t)sa * R.total + R.entry_base[a];
-            uint32_t* crb = cr + (uint64_t)sb * R.total + R.entry_base[b];
-
             for (uint32_t pos = d; pos < dend; pos++) {
                 bool take_a;
                 if (ja >= lenA)
@@ -464,7 +464,7 @@ __global__ __launch_bounds__(CORANK_BLOCK) void k_corank(
                             cmp_keys_aux(R, aux, a, iaS + ja, b, gib) == 0)
                             v |= CR_LOSER;
                     }
-                    cra[iaS + ja] = v;
+                    s_cr[ja] = v;
                     ja++;
                 } else {
                     uint64_t gia = iaS + ja;
@@ -476,10 +476,24 @@ __global__ __launch_bounds__(CORANK_BLOCK) void k_corank(
                             cmp_keys_aux(R, aux, b, ibS + jb, a, gia) == 0)
                             v |= CR_LOSER;
                     }
-                    crb[ibS + jb] = v;
+                    s_cr[lenA + jb] = v;
                     jb++;
                 }
             }
+        }
+        __syncthreads();
+        /* coalesced writeback of both contiguous output ranges */
+        {
+            uint32_t sa = (uint32_t)(b < a ? b : b - 1); /* slot of b in a */
+            uint32_t sb = (uint32_t)(a < b ? a : a - 1); /* slot of a in b */
+            uint32_t* cra = cr + (uint64_t)sa * R.total + R.entry_base[a] +
+                            iaS;
+            uint32_t* crb = cr + (uint64_t)sb * R.total + R.entry_base[b] +
+                            ibS;
+            for (uint32_t u = threadIdx.x; u < lenA; u += CORANK_BLOCK)
+                cra[u] = s_cr[u];
+            for (uint32_t u = threadIdx.x; u < lenB; u += CORANK_BLOCK)
+                crb[u] = s_cr[lenA + u];
         }
         __syncthreads();
     }
