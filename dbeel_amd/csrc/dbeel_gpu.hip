@@ -604,16 +604,20 @@ struct RankRecFlag {
 #define COPY_SPAN 516
 
 #define COPY_GRANULES (COPY_WINDOW / 16)
+/* geometry variants for within-probe A/B (env DBEEL_COPY_VARIANT):
+ * 0 = 256 threads x 16 KiB (default), 1 = 256 x 8 KiB,
+ * 2 = 512 x 16 KiB, 3 = 512 x 32 KiB */
 
 /* Per destination window, the largest survivor p with offset(p) <= window
  * start — one thread per window (parallel), consumed by k_copy. */
 __global__ void k_winmap(const uint8_t* out_index, uint64_t n_surv,
-                         uint64_t total_bytes, uint32_t* win_p0) {
-    uint64_t n_windows = (total_bytes + COPY_WINDOW - 1) / COPY_WINDOW;
+                         uint64_t total_bytes, uint32_t win,
+                         uint32_t* win_p0) {
+    uint64_t n_windows = (total_bytes + win - 1) / win;
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
     for (uint64_t w = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
          w < n_windows; w += stride) {
-        uint64_t wstart = w * COPY_WINDOW;
+        uint64_t wstart = (uint64_t)w * win;
         uint64_t lo = 0, hi = n_surv; /* first offset > wstart, minus 1 */
         while (lo < hi) {
             uint64_t mid = (lo + hi) >> 1;
@@ -626,17 +630,20 @@ __global__ void k_winmap(const uint8_t* out_index, uint64_t n_surv,
     }
 }
 
-__global__ __launch_bounds__(COPY_BLOCK) void k_copy(
+template <int BLK, int WIN>
+__global__ __launch_bounds__(BLK) void k_copy(
     const uint8_t* out_index, const uint64_t* src_map,
     const uint32_t* win_p0, uint64_t n_surv, uint64_t total_bytes,
     uint8_t* out_data) {
-    __shared__ uint64_t s_off[COPY_SPAN + 1];
-    __shared__ uint64_t s_src[COPY_SPAN];
-    __shared__ uint16_t s_gid[COPY_GRANULES]; /* granule -> entry slot */
+    constexpr int SPAN = WIN / 32 + 4;
+    constexpr int GRAN = WIN / 16;
+    __shared__ uint64_t s_off[SPAN + 1];
+    __shared__ uint64_t s_src[SPAN];
+    __shared__ uint16_t s_gid[GRAN]; /* granule -> entry slot */
 
-    uint64_t n_windows = (total_bytes + COPY_WINDOW - 1) / COPY_WINDOW;
+    uint64_t n_windows = (total_bytes + WIN - 1) / WIN;
     for (uint64_t w = blockIdx.x; w < n_windows; w += gridDim.x) {
-        uint64_t wstart = w * COPY_WINDOW;
+        uint64_t wstart = w * WIN;
         uint64_t p0 = win_p0[w];
         /* entries intersecting this window are p0 .. win_p0[w+1]; staging
          * only those (not the worst-case SPAN) saves ~30x index re-reads
@@ -644,9 +651,9 @@ __global__ __launch_bounds__(COPY_BLOCK) void k_copy(
         uint64_t p_end = (w + 1 < n_windows) ? (uint64_t)win_p0[w + 1] + 1
                                              : n_surv;
         uint32_t cnt = (uint32_t)(p_end - p0);
-        if (cnt > COPY_SPAN) cnt = COPY_SPAN;
+        if (cnt > SPAN) cnt = SPAN;
         if (cnt > n_surv - p0) cnt = (uint32_t)(n_surv - p0);
-        for (uint32_t u = threadIdx.x; u <= cnt; u += COPY_BLOCK) {
+        for (uint32_t u = threadIdx.x; u <= cnt; u += BLK) {
             uint64_t p = p0 + u;
             if (u == cnt)
                 s_off[cnt] = (p < n_surv) ? ld_u64(out_index + p * 16)
@@ -659,18 +666,18 @@ __global__ __launch_bounds__(COPY_BLOCK) void k_copy(
         __syncthreads();
         /* fill the granule -> entry map: entry slot u owns granules whose
          * START byte lies in [s_off[u], s_off[u+1]) */
-        for (uint32_t u = threadIdx.x; u < cnt; u += COPY_BLOCK) {
+        for (uint32_t u = threadIdx.x; u < cnt; u += BLK) {
             uint64_t b0 = s_off[u], b1 = s_off[u + 1];
             uint64_t g0 = (b0 <= wstart) ? 0 : ((b0 - wstart + 15) >> 4);
             uint64_t g1 = (b1 - wstart + 15) >> 4; /* exclusive */
-            if (g1 > COPY_GRANULES) g1 = COPY_GRANULES;
+            if (g1 > GRAN) g1 = GRAN;
             for (uint64_t g = g0; g < g1; g++) s_gid[g] = (uint16_t)u;
         }
         __syncthreads();
 
         #pragma unroll
-        for (int q = 0; q < COPY_GRANULES / COPY_BLOCK; q++) {
-            uint32_t gl = q * COPY_BLOCK + threadIdx.x; /* window-local */
+        for (int q = 0; q < GRAN / BLK; q++) {
+            uint32_t gl = q * BLK + threadIdx.x; /* window-local */
             uint64_t gpos = wstart + (uint64_t)gl * 16;
             if (gpos >= total_bytes) break;
             uint32_t j = s_gid[gl];
@@ -1054,8 +1061,9 @@ extern "C" int dbeel_gpu_job_create(const dbeel_run_view* runs, size_t n_runs,
     JOB_CHECK(hipMalloc(&job->d_aux, n * 64));
     JOB_CHECK(hipMalloc(&job->d_cr,
                         (n_runs > 1 ? (n_runs - 1) * n : 1) * 4));
+    /* sized for the smallest copy-window variant (8 KiB) */
     JOB_CHECK(hipMalloc(&job->d_winp0,
-                        (total_data / COPY_WINDOW + 2) * sizeof(uint32_t)));
+                        (total_data / 8192 + 2) * sizeof(uint32_t)));
 
     size_t t1 = 0, t2 = 0;
     rocprim::exclusive_scan(nullptr, t1,
@@ -1264,12 +1272,23 @@ extern "C" int dbeel_gpu_job_run(dbeel_gpu_job* job, int keep_tombstones,
 
     HIP_CHECK(hipEventRecord(job->ev[4], s));
     if (total_out) {
-        uint64_t windows = (total_out + COPY_WINDOW - 1) / COPY_WINDOW;
+        int variant = 0;
+        if (const char* v = getenv("DBEEL_COPY_VARIANT")) variant = atoi(v);
+        uint32_t win = (variant == 1) ? 8192
+                       : (variant == 3) ? 32768
+                                        : 16384;
+        uint32_t blk = (variant >= 2) ? 512 : 256;
+        uint64_t windows = (total_out + win - 1) / win;
         uint32_t grid = windows > 4096 ? 4096 : (uint32_t)windows;
         hipLaunchKernelGGL(k_winmap, dim3(pick_grid(windows, 256)), dim3(256),
-                           0, s, job->d_outindex, n_surv, total_out,
+                           0, s, job->d_outindex, n_surv, total_out, win,
                            job->d_winp0);
-        hipLaunchKernelGGL(k_copy, dim3(grid), dim3(COPY_BLOCK), 0, s,
+        void (*kc)(const uint8_t*, const uint64_t*, const uint32_t*,
+                   uint64_t, uint64_t, uint8_t*) = k_copy<256, 16384>;
+        if (variant == 1) kc = k_copy<256, 8192>;
+        if (variant == 2) kc = k_copy<512, 16384>;
+        if (variant == 3) kc = k_copy<512, 32768>;
+        hipLaunchKernelGGL(kc, dim3(grid), dim3(blk), 0, s,
                            job->d_outindex, job->d_srcmap, job->d_winp0,
                            n_surv, total_out, job->d_outdata);
     }
