@@ -335,7 +335,7 @@ __global__ void k_prepare(RunsDesc R, uint64_t* pfx, Aux* aux,
 /* ------------------------------------------------------------------ */
 
 #define CORANK_BLOCK 256
-#define CORANK_STEPS 16
+#define CORANK_STEPS 8
 #define CORANK_BLOCK_POS (CORANK_BLOCK * CORANK_STEPS) /* 4096 */
 #define CR_LOSER 0x80000000u
 #define CR_MASK 0x7FFFFFFFu
