@@ -67,8 +67,10 @@ def main():
                     help="cfg3 = 8-run x 1 GiB metric config; cfg4 = 8 "
                          "independent 4-run x 256 MiB jobs per GPU")
     ap.add_argument("--keep-tombstones", action="store_true")
-    ap.add_argument("--cpu-baseline-scale", type=float, default=0.25,
-                    help="fraction of the workload timed on 1 host core")
+    ap.add_argument("--cpu-baseline-scale", type=float, default=1.0,
+                    help="fraction of the workload timed on the host "
+                         "(default: the whole workload, one single-threaded "
+                         "oracle job per core)")
     ap.add_argument("--skip-cpu-baseline", action="store_true")
     args = ap.parse_args()
 
