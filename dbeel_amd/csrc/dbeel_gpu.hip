@@ -13,7 +13,8 @@
  *   k_corank     : block-cooperative merge-path co-ranking — for every
  *                  run pair, diagonal-partitioned 4096-position windows;
  *                  blocks stage both prefix segments into LDS coalesced
- *                  and walk 16 merged positions per thread, recording
+ *                  and walk CORANK_STEPS merged positions per thread,
+ *                  recording
  *                  crossranks (order = key bytes asc, timestamp i128 asc,
  *                  run index asc — lsm_tree.rs:52-71 + mod.rs:75-81) and
  *                  newest-wins supersession flags (lsm_tree.rs:1041-1044).
@@ -25,8 +26,9 @@
  *   k_emit       : output .index records (offset/key_size/full_size,
  *                  entry_writer.rs:79-87) + compacted source map.
  *   k_winmap/k_copy : verbatim survivor copy, balanced by DESTINATION
- *                  granule (4 x 16-B granules per lane, 16-KiB window per
- *                  256-thread block, LDS granule->entry map) so throughput
+ *                  granule (16-B granules per lane, 16-KiB window per
+ *                  256-thread block by default, LDS granule->entry map;
+ *                  geometry variants for A/B) so throughput
  *                  is independent of entry size; aligned non-temporal
  *                  16-B stores, unaligned 16-B loads.
  *   k_encode_*   : the memtable-flush run encoder (lsm_tree.rs:925-946).
@@ -265,8 +267,8 @@ __device__ __forceinline__ int cmp_aux_full(const RunsDesc& R,
 
 /* Validates every entry (bounds + bincode field cross-check,
  * read_next_entry lsm_tree.rs:1158-70) and extracts the dense key-prefix
- * and aux arrays the rank searches run on. Run sortedness is checked in
- * k_rank (it already holds consecutive entries). */
+ * and aux arrays the merge runs on. Run sortedness is checked in
+ * k_rankreduce (on the dense pfx/aux arrays). */
 __global__ void k_prepare(RunsDesc R, uint64_t* pfx, Aux* aux,
                           uint32_t* err) {
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
@@ -320,7 +322,7 @@ __global__ void k_prepare(RunsDesc R, uint64_t* pfx, Aux* aux,
 /* partitioned into CORANK_BLOCK_POS-wide windows by diagonal binary   */
 /* search. Each 256-thread block stages the two pfx segments of its    */
 /* window into LDS with coalesced loads, sub-partitions the window     */
-/* 16 merged positions per thread (diagonal search inside LDS), and    */
+/* CORANK_STEPS merged positions per thread (diagonal search in LDS),  */
 /* walks linearly, recording for every consumed entry its crossrank    */
 /* into the opposite run (= the opposite cursor) and whether the       */
 /* opposite run's next entry carries the same key (then it is later    */
