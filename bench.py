@@ -34,6 +34,15 @@ def make_workload(rank: int, scale: float, workload: str = "cfg3"):
     64 jobs over 8 GPUs, one dbeel shard each). Returns (job_list, cfg)."""
     from dbeel_amd.genruns import CONFIGS, make_runs
 
+    if workload == "cfg5":
+        from dbeel_amd.genruns import make_runs_varkey
+
+        cfg = dict(CONFIGS["cfg5"])
+        if scale != 1.0:
+            cfg["entries_per_run"] = max(64, int(cfg["entries_per_run"] * scale))
+        runs = make_runs_varkey(seed=0xDBEE1 + 7919 * rank, **cfg)
+        cfg["key_size"] = 128  # upper bound; keys are 8-128 B zipf
+        return [runs], cfg
     if workload == "cfg4":
         cfg = dict(CONFIGS["cfg4_job"])
         if scale != 1.0:
@@ -63,9 +72,11 @@ def main():
     ap.add_argument("--warmup", type=int, default=3)
     ap.add_argument("--scale", type=float, default=1.0,
                     help="scale entries_per_run (1.0 = full config 3)")
-    ap.add_argument("--workload", choices=["cfg3", "cfg4"], default="cfg3",
+    ap.add_argument("--workload", choices=["cfg3", "cfg4", "cfg5"],
+                    default="cfg3",
                     help="cfg3 = 8-run x 1 GiB metric config; cfg4 = 8 "
-                         "independent 4-run x 256 MiB jobs per GPU")
+                         "independent 4-run x 256 MiB jobs per GPU; cfg5 = "
+                         "16-run var-length msgpack keys + 4 KiB values")
     ap.add_argument("--keep-tombstones", action="store_true")
     ap.add_argument("--cpu-baseline-scale", type=float, default=1.0,
                     help="fraction of the workload timed on the host "
@@ -276,8 +287,9 @@ def main():
             "dtype": "u8",
             "data": "synthetic",
             "config": {
-                "workload": ("cfg3_8run_x_1GiB" if args.workload == "cfg3"
-                             else "cfg4_8jobs_4run_x_256MiB_per_gpu")
+                "workload": {"cfg3": "cfg3_8run_x_1GiB",
+                             "cfg4": "cfg4_8jobs_4run_x_256MiB_per_gpu",
+                             "cfg5": "cfg5_16run_varkey_4KiB"}[args.workload]
                 + (f"_scale{args.scale}" if args.scale != 1.0 else ""),
                 "n_runs": cfg["n_runs"],
                 "entries_per_run": cfg["entries_per_run"],

@@ -8,7 +8,7 @@ Shapes (BASELINE.json `configs`):
         5% tombstones
   cfg4: 64 jobs of (4 runs x 256 MiB), cfg2 shapes — built per-job via
         make_job(..., job_seed)
-  cfg5: variable-length keys (later round)
+  cfg5: 16 runs, zipf msgpack str keys (8-128 B), 4 KiB values
 
 Determinism: default seed 0xDBEE1; timestamps strictly increase with run
 index (ts = run_index * 2^40 + position) so merge winners are deterministic
@@ -124,6 +124,11 @@ CONFIGS = {
     "cfg4_job": dict(
         n_runs=4, entries_per_run=883_000, key_size=16, value_size=256
     ),
+    # cfg5: built via make_runs_varkey (variable-length msgpack keys)
+    "cfg5": dict(
+        n_runs=16, entries_per_run=200_000, value_size=4096,
+        overlap_frac=0.3, tombstone_frac=0.02
+    ),
 }
 
 
@@ -149,19 +154,23 @@ def make_runs_varkey(
     """cfg5-shaped runs: variable-length msgpack str keys, zipf(1.1)
     lengths clipped to 8..128 B total, 4 KiB values (BASELINE.json
     configs[4]). Keys are raw msgpack `str8` encodings (0xd9 | len |
-    utf8-ish bytes) compared as raw bytes, exactly as dbeel compares them
-    (Entry::cmp on Vec<u8>, mod.rs:75-81). Python-loop builder — parity
-    test scale, not bench scale."""
-    from .format import Entry, build_run
-
+    bytes) compared as raw bytes, exactly as dbeel compares them
+    (Entry::cmp on Vec<u8>, mod.rs:75-81). Bulk-RNG path: ~4 s per
+    1 GiB run."""
     rng = np.random.default_rng(seed)
 
     def draw_keys(n):
         keys = set()
         while len(keys) < n:
-            L = int(np.clip(rng.zipf(1.1), 6, 126))
-            body = bytes(rng.integers(32, 127, L, dtype=np.uint8))
-            keys.add(bytes([0xD9, L]) + body)
+            need = n - len(keys) + 16
+            Ls = np.clip(rng.zipf(1.1, need), 6, 126).astype(np.int64)
+            body = rng.integers(32, 127, int(Ls.sum()), dtype=np.uint8)
+            pos = 0
+            for L in Ls:
+                keys.add(bytes([0xD9, L]) + body[pos : pos + L].tobytes())
+                pos += L
+                if len(keys) >= n:
+                    break
         return list(keys)
 
     n_shared = int(entries_per_run * overlap_frac)
@@ -180,12 +189,55 @@ def make_runs_varkey(
                 spos += n_shared
             ks = own + sh[: n_shared]
         ks = sorted(set(ks))
-        ents = []
-        for i, k in enumerate(ks):
-            tomb = rng.random() < tombstone_frac
-            data = b"" if tomb else bytes(
-                rng.integers(0, 256, value_size, dtype=np.uint8)
-            )
-            ents.append(Entry(k, data, (r << 40) + i))
-        out.append(build_run(ents))
+        n = len(ks)
+        tomb = rng.random(n) < tombstone_frac
+        n_norm = int(n - tomb.sum())
+        vblob = rng.integers(0, 256, n_norm * value_size,
+                             dtype=np.uint8).tobytes()
+        out.append(_build_run_var(ks, tomb, vblob, value_size, r << 40))
     return out
+
+
+def _build_run_var(ks, tomb, vblob, V, ts_base):
+    """Fast var-key run builder: one join over per-field pieces
+    (memoryview value slices — no per-entry intermediate copies)."""
+    import struct
+
+    from .format import INDEX_DTYPE
+
+    n = len(ks)
+    idx = np.zeros(n, dtype=INDEX_DTYPE)
+    parts = []
+    mv = memoryview(vblob)
+    dlenV = struct.pack("<Q", V)
+    dlen0 = struct.pack("<Q", 0)
+    klen_cache = {}
+    off = 0
+    vpos = 0
+    offs = idx["offset"]
+    kss = idx["key_size"]
+    fss = idx["full_size"]
+    for i, k in enumerate(ks):
+        kl = len(k)
+        hdr = klen_cache.get(kl)
+        if hdr is None:
+            hdr = struct.pack("<Q", kl)
+            klen_cache[kl] = hdr
+        parts.append(hdr)
+        parts.append(k)
+        if tomb[i]:
+            parts.append(dlen0)
+            dl = 0
+        else:
+            parts.append(dlenV)
+            parts.append(mv[vpos : vpos + V])
+            vpos += V
+            dl = V
+        parts.append(int(ts_base + i).to_bytes(16, "little"))
+        offs[i] = off
+        kss[i] = 8 + kl
+        fss[i] = 32 + kl + dl
+        off += 32 + kl + dl
+    data = b"".join(parts)
+    return (np.frombuffer(data, dtype=np.uint8),
+            idx.view(np.uint8).reshape(-1))
