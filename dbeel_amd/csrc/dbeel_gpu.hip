@@ -632,7 +632,7 @@ __global__ void k_winmap(const uint8_t* out_index, uint64_t n_surv,
     }
 }
 
-template <int BLK, int WIN>
+template <int BLK, int WIN, bool FUNNEL = true>
 __global__ __launch_bounds__(BLK) void k_copy(
     const uint8_t* out_index, const uint64_t* src_map,
     const uint32_t* win_p0, uint64_t n_surv, uint64_t total_bytes,
@@ -716,7 +716,8 @@ __global__ __launch_bounds__(BLK) void k_copy(
                     typedef unsigned int v4u
                         __attribute__((ext_vector_type(4)));
                     v4u v;
-                    uint32_t sh = (uint32_t)((uintptr_t)src & 15);
+                    uint32_t sh =
+                        FUNNEL ? (uint32_t)((uintptr_t)src & 15) : 0u;
                     if (sh == 0) {
                         __builtin_memcpy(&v, src, 16);
                     } else {

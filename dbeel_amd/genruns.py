@@ -171,7 +171,11 @@ def make_runs_varkey(
                 pos += L
                 if len(keys) >= n:
                     break
-        return list(keys)
+        # set iteration order is hash-salted per process; sort for
+        # determinism, then rng-permute so the shared-key slices below
+        # remain randomly selected (not contiguous key ranges)
+        lst = sorted(keys)
+        return [lst[i] for i in rng.permutation(len(lst))]
 
     n_shared = int(entries_per_run * overlap_frac)
     shared = draw_keys(n_shared * (n_runs // 2)) if n_shared else []
