@@ -155,8 +155,7 @@ def make_runs_varkey(
     lengths clipped to 8..128 B total, 4 KiB values (BASELINE.json
     configs[4]). Keys are raw msgpack `str8` encodings (0xd9 | len |
     bytes) compared as raw bytes, exactly as dbeel compares them
-    (Entry::cmp on Vec<u8>, mod.rs:75-81). Bulk-RNG path: ~4 s per
-    1 GiB run."""
+    (Entry::cmp on Vec<u8>, mod.rs:75-81)."""
     rng = np.random.default_rng(seed)
 
     def draw_keys(n):
