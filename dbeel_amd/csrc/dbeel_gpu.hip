@@ -632,7 +632,7 @@ __global__ void k_winmap(const uint8_t* out_index, uint64_t n_surv,
     }
 }
 
-template <int BLK, int WIN, bool FUNNEL = true>
+template <int BLK, int WIN>
 __global__ __launch_bounds__(BLK) void k_copy(
     const uint8_t* out_index, const uint64_t* src_map,
     const uint32_t* win_p0, uint64_t n_surv, uint64_t total_bytes,
@@ -716,52 +716,54 @@ __global__ __launch_bounds__(BLK) void k_copy(
                     typedef unsigned int v4u
                         __attribute__((ext_vector_type(4)));
                     v4u v;
-                    uint32_t sh =
-                        FUNNEL ? (uint32_t)((uintptr_t)src & 15) : 0u;
-                    if (sh == 0) {
-                        __builtin_memcpy(&v, src, 16);
-                    } else {
-                        /* misaligned source (odd entry sizes): two ALIGNED
-                         * 16-B loads + byte funnel shift — a misaligned
-                         * load is split by the hardware and halves the
-                         * issue rate (measured 2x on 4 KiB values). The
-                         * input slab is padded +-16 B for the base. */
-                        const uint8_t* ab = src - sh;
-                        uint64_t w[4];
-                        __builtin_memcpy(&w[0], ab, 16);
-                        __builtin_memcpy(&w[2], ab + 16, 16);
-                        uint32_t k = sh & 7;
-                        uint64_t a0, a1, a2;
-                        if (sh < 8) {
-                            a0 = w[0];
-                            a1 = w[1];
-                            a2 = w[2];
-                        } else {
-                            a0 = w[1];
-                            a1 = w[2];
-                            a2 = w[3];
-                        }
-                        uint64_t lo, hi;
-                        if (k == 0) {
-                            lo = a0;
-                            hi = a1;
-                        } else {
-                            lo = (a0 >> (8 * k)) | (a1 << (64 - 8 * k));
-                            hi = (a1 >> (8 * k)) | (a2 << (64 - 8 * k));
-                        }
-                        __builtin_memcpy(&v, &lo, 8);
-                        __builtin_memcpy(reinterpret_cast<uint8_t*>(&v) + 8,
-                                         &hi, 8);
-                    }
+                    __builtin_memcpy(&v, src, 16);
                     /* streamed once, never re-read: keep L2 for sources */
                     __builtin_nontemporal_store(
                         v, reinterpret_cast<v4u*>(dst));
                 } else {
                     for (uint32_t b = 0; b < nbytes; b++) dst[b] = src[b];
                 }
+            } else if (nbytes == 16) {
+                /* one entry boundary inside the granule: 16-B blend of the
+                 * entry tail and the next survivor's head. Byte loops here
+                 * cost a whole wave ~3k serial cycles, and odd entry sizes
+                 * put one straddle in nearly every wave (measured 2.4x on
+                 * odd-size values). Reading past either entry stays inside
+                 * the padded input slab. */
+                uint32_t c1 = (uint32_t)(e_end - gpos); /* 1..15 */
+                const uint8_t* src2 = (const uint8_t*)s_src[j + 1];
+                uint64_t l0, h0, l2, h2;
+                __builtin_memcpy(&l0, src, 8);
+                __builtin_memcpy(&h0, src + 8, 8);
+                __builtin_memcpy(&l2, src2, 8);
+                __builtin_memcpy(&h2, src2 + 8, 8);
+                uint64_t lo, hi;
+                if (c1 < 8) {
+                    uint32_t s = 8 * c1;
+                    uint64_t m = (~0ull) >> (64 - s);
+                    lo = (l0 & m) | (l2 << s);
+                    hi = (h2 << s) | (l2 >> (64 - s));
+                } else {
+                    uint32_t c = c1 - 8; /* 0..7 */
+                    lo = l0;
+                    if (c == 0) {
+                        hi = l2;
+                    } else {
+                        uint64_t m = (~0ull) >> (64 - 8 * c);
+                        hi = (h0 & m) | (l2 << (8 * c));
+                    }
+                }
+                typedef unsigned int v4u
+                    __attribute__((ext_vector_type(4)));
+                v4u v;
+                __builtin_memcpy(&v, &lo, 8);
+                __builtin_memcpy(reinterpret_cast<uint8_t*>(&v) + 8, &hi, 8);
+                __builtin_nontemporal_store(v,
+                                            reinterpret_cast<v4u*>(dst));
             } else {
-                /* one entry boundary inside the granule */
+                /* output tail granule (< 16 B, at most one per job) */
                 uint32_t c1 = (uint32_t)(e_end - gpos);
+                if (c1 > nbytes) c1 = nbytes;
                 for (uint32_t b = 0; b < c1; b++) dst[b] = src[b];
                 const uint8_t* src2 = (const uint8_t*)s_src[j + 1];
                 for (uint32_t b = c1; b < nbytes; b++)
