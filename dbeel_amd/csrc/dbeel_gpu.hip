@@ -1048,10 +1048,20 @@ static int validate_runs(const dbeel_run_view* runs, size_t n_runs) {
             set_err("run %zu: data_len too large", r);
             return DBEEL_ERR_ITEM_TOO_LARGE;
         }
-        if (runs[r].index_len / 16 >= (1ull << 32)) {
-            set_err("run %zu: more than 2^32 entries unsupported", r);
+        if (runs[r].index_len / 16 >= (1ull << 31)) {
+            /* crossranks carry a 31-bit payload (CR_LOSER uses bit 31):
+             * enforce the representation bound explicitly rather than by
+             * incidental OOM */
+            set_err("run %zu: 2^31 or more entries unsupported", r);
             return DBEEL_ERR_ITEM_TOO_LARGE;
         }
+    }
+    uint64_t total = 0;
+    for (size_t r = 0; r < n_runs; r++) total += runs[r].index_len / 16;
+    if (total >= (1ull << 32)) {
+        /* survivor positions (d_pos scan, win_p0) are uint32 */
+        set_err("job has 2^32 or more total entries");
+        return DBEEL_ERR_ITEM_TOO_LARGE;
     }
     return DBEEL_OK;
 }
@@ -1520,12 +1530,14 @@ extern "C" int dbeel_gpu_job_fetch(dbeel_gpu_job* job,
 /*                                                                    */
 /* GPU analogue of LSMTree::get over the sstables                     */
 /* (lsm_tree.rs:605-723, binary search per run lsm_tree.rs:674-723):  */
-/* one thread per query key binary-searches every run and keeps the   */
-/* newest match — max (timestamp, run index), the same winner rule as */
-/* compaction. Tombstones are reported as found with value_len 0 so   */
-/* the caller distinguishes deleted from absent (tests/db_server.rs   */
-/* delete->get->KeyNotFound semantics). Bloom prefiltering stays on   */
-/* the host (dbeel_bloom_contains).                                   */
+/* one thread per query key binary-searches the runs newest-INDEX-    */
+/* first and returns the first key match, exactly like the reference  */
+/* read path (`sstables.iter().rev()`, lsm_tree.rs:692-696) — the     */
+/* highest run index wins regardless of timestamp. Tombstones are     */
+/* reported as found with value_len 0 so the caller distinguishes     */
+/* deleted from absent (tests/db_server.rs delete->get->KeyNotFound   */
+/* semantics). Bloom prefiltering stays on the host                   */
+/* (dbeel_bloom_contains).                                            */
 /* ------------------------------------------------------------------ */
 
 __global__ void k_lookup(RunsDesc R, const uint8_t* keys,
@@ -1538,9 +1550,8 @@ __global__ void k_lookup(RunsDesc R, const uint8_t* keys,
         uint64_t klen = key_off[q + 1] - key_off[q];
         int best_run = -1;
         uint64_t best_off = 0, best_vlen = 0;
-        uint64_t best_tslo = 0;
-        int64_t best_tshi = 0;
-        for (int r = 0; r < R.n_runs; r++) {
+        /* newest-index-first, first match wins (lsm_tree.rs:692-696) */
+        for (int r = R.n_runs - 1; r >= 0; r--) {
             uint64_t lo = 0, hi = R.count[r];
             while (lo < hi) {
                 uint64_t mid = (lo + hi) >> 1;
@@ -1560,19 +1571,10 @@ __global__ void k_lookup(RunsDesc R, const uint8_t* keys,
             if (!load_entry(R, r, lo, m)) continue;
             if (m.klen != klen || cmp_keys(m.key, m.klen, key, klen) != 0)
                 continue;
-            uint64_t tslo;
-            int64_t tshi;
-            load_ts(m, tslo, tshi);
-            /* newest wins: (timestamp, run index); runs scan ascending so
-             * a later run with equal ts wins automatically */
-            if (best_run < 0 || tshi > best_tshi ||
-                (tshi == best_tshi && tslo >= best_tslo)) {
-                best_run = r;
-                best_off = m.off + 16 + m.klen; /* value bytes start */
-                best_vlen = (uint64_t)m.full_size - 32 - m.klen;
-                best_tslo = tslo;
-                best_tshi = tshi;
-            }
+            best_run = r;
+            best_off = m.off + 16 + m.klen; /* value bytes start */
+            best_vlen = (uint64_t)m.full_size - 32 - m.klen;
+            break;
         }
         out[q].run = best_run;
         out[q].is_tombstone = (best_run >= 0 && best_vlen == 0) ? 1 : 0;

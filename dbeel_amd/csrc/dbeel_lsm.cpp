@@ -198,7 +198,7 @@ extern "C" int dbeel_lsm_replay(const char* dir_c, uint32_t* out_replayed) {
             for (auto& rn : renames)
                 if (exists(rn.first))
                     if (rename(rn.first.c_str(), rn.second.c_str()) != 0)
-                        return DBEEL_ERR_HIP; /* io-class failure */
+                        return DBEEL_ERR_IO;
         }
         unlink(jpath.c_str());
         replayed++;
@@ -223,7 +223,7 @@ extern "C" int dbeel_lsm_compact(const char* dir_c, const uint64_t* indices,
     for (size_t i = 0; i < n_indices; i++) {
         if (!read_whole(file_path(dir, indices[i], "data"), datas[i]) ||
             !read_whole(file_path(dir, indices[i], "index"), idxs[i]))
-            return DBEEL_ERR_INVALID_ARG;
+            return DBEEL_ERR_IO;
         views[i] = {datas[i].data(), datas[i].size(), idxs[i].data(),
                     idxs[i].size()};
         total_input_data += datas[i].size();
@@ -260,7 +260,7 @@ extern "C" int dbeel_lsm_compact(const char* dir_c, const uint64_t* indices,
     uint64_t written = res.entries_written;
     dbeel_gpu_result_free(&res);
     dbeel_gpu_bloom_free(bloom);
-    if (!ok) return DBEEL_ERR_HIP;
+    if (!ok) return DBEEL_ERR_IO;
 
     /* journal (CompactionAction, lsm_tree.rs:1090-1105): renames of all
      * three staging files, deletes of every input's data/index/bloom */
@@ -282,7 +282,7 @@ extern "C" int dbeel_lsm_compact(const char* dir_c, const uint64_t* indices,
     for (auto& d : deletes) put_str(journal, d);
     std::string jpath = file_path(dir, output_index, "compact_action");
     if (!write_whole(jpath, journal.data(), journal.size()))
-        return DBEEL_ERR_HIP;
+        return DBEEL_ERR_IO;
 
     /* crash-injection hook for recovery tests (the flow_events analogue,
      * reference flow_events.rs:7-14): stop after the journal is durable —
@@ -302,7 +302,7 @@ extern "C" int dbeel_lsm_compact(const char* dir_c, const uint64_t* indices,
     for (auto& rn : renames)
         if (exists(rn.from))
             if (rename(rn.from.c_str(), rn.to.c_str()) != 0)
-                return DBEEL_ERR_HIP;
+                return DBEEL_ERR_IO;
     for (auto& d : deletes)
         if (exists(d)) unlink(d.c_str());
     unlink(jpath.c_str());

@@ -22,6 +22,7 @@ ERROR_NAMES = {
     4: "HIP",
     5: "NO_GPU",
     6: "OOM",
+    7: "IO",
 }
 
 
@@ -184,7 +185,9 @@ class LookupHit(ctypes.Structure):
 def lookup(runs, keys, device: int = 0):
     """Batched point lookup (LSMTree::get over sstables,
     lsm_tree.rs:605-723): for each key returns the value bytes of the
-    newest match, b"" for a deleted key (tombstone), or None if absent."""
+    first match scanning runs newest-index-first (the reference's
+    `sstables.iter().rev()`, lsm_tree.rs:692-696 — highest run index
+    wins), b"" for a deleted key (tombstone), or None if absent."""
     lib = load()
     if not hasattr(lib, "_lookup_ready"):
         lib.dbeel_gpu_lookup.restype = ctypes.c_int

@@ -87,9 +87,11 @@ enum {
     DBEEL_ERR_INVALID_ARG  = 1,
     DBEEL_ERR_CORRUPT      = 2,  /* bincode-decode-failure analogue */
     DBEEL_ERR_ITEM_TOO_LARGE = 3,
-    DBEEL_ERR_HIP          = 4,  /* device/runtime failure (I/O analogue) */
+    DBEEL_ERR_HIP          = 4,  /* device/runtime failure */
     DBEEL_ERR_NO_GPU       = 5,
     DBEEL_ERR_OOM          = 6,
+    DBEEL_ERR_IO           = 7,  /* host filesystem failure (error.rs I/O
+                                    analogue) — distinct from device errors */
 };
 
 /* Compact n_runs runs into one run. keep_tombstones: 0 = drop entries with
@@ -114,9 +116,13 @@ void dbeel_gpu_result_free(dbeel_compact_result* r);
 const char* dbeel_gpu_last_error(void);
 
 /* ---- Batched point lookup ----
- * GPU analogue of LSMTree::get over the sstables (lsm_tree.rs:605-723):
- * for each query key, the newest match across the runs — max (timestamp,
- * run index), the compaction winner rule. run = -1 when absent;
+ * GPU analogue of LSMTree::get over the sstables (lsm_tree.rs:674-723):
+ * the reference scans sstables newest-INDEX-first and returns the first
+ * key match (`sstables.iter().rev()`, lsm_tree.rs:692-696) — so the match
+ * in the HIGHEST run index wins, regardless of timestamp. (This can
+ * differ from the compaction winner rule when set_with_timestamp writes
+ * an older timestamp into a newer sstable; compaction then reorders —
+ * faithful to the reference read path.) run = -1 when absent;
  * is_tombstone = 1 distinguishes deleted from absent (the reference's
  * delete->get->KeyNotFound semantics). value_offset indexes into
  * runs[run].data. Bloom prefiltering stays host-side

@@ -72,6 +72,42 @@ def test_device_minus_one_rejected():
     assert ei.value.code == 1  # INVALID_ARG
 
 
+def test_entry_count_bounds_rejected():
+    """Crossranks carry a 31-bit payload (CR_LOSER flag in bit 31), so a
+    run with >= 2^31 entries must be rejected up front —
+    validation runs before any device access, so this tests on CPU. The
+    fake index pointer is never dereferenced: the length checks fire
+    first."""
+    _built()
+    import numpy as np
+
+    from dbeel_amd.engine import RunView, load
+
+    lib = load()
+    buf = np.zeros(16, dtype=np.uint8)
+    views = (RunView * 1)()
+    views[0].data = buf.ctypes.data_as(
+        ctypes.POINTER(ctypes.c_uint8))
+    views[0].data_len = 16
+    views[0].index = buf.ctypes.data_as(
+        ctypes.POINTER(ctypes.c_uint8))
+    views[0].index_len = (1 << 31) * 16  # 2^31 entries
+    out = ctypes.c_void_p()
+    rc = lib.dbeel_gpu_job_create(views, 1, 0, ctypes.byref(out))
+    assert rc == 3, rc  # ITEM_TOO_LARGE
+
+    # per-run counts under 2^31 but job total >= 2^32 entries: the u32
+    # survivor positions (d_pos, win_p0) bound the JOB, also rejected
+    views2 = (RunView * 3)()
+    for v in views2:
+        v.data = buf.ctypes.data_as(ctypes.POINTER(ctypes.c_uint8))
+        v.data_len = 16
+        v.index = buf.ctypes.data_as(ctypes.POINTER(ctypes.c_uint8))
+        v.index_len = ((1 << 31) - 1) * 16  # sums past 2^32
+    rc = lib.dbeel_gpu_job_create(views2, 3, 0, ctypes.byref(out))
+    assert rc == 3, rc
+
+
 def test_no_gpu_errors_loudly():
     """On a GPU-less host, using a device ordinal must raise, never silently
     fall back."""

@@ -215,9 +215,11 @@ def test_encode_run_matches_entry_writer_layout(engine):
 
 
 def test_batched_lookup(engine):
-    """GPU-batched point lookup vs a host-side
-    newest-wins model: present, overwritten, deleted and absent keys,
-    plus the delete->get->KeyNotFound distinction
+    """GPU-batched point lookup vs a host-side model of
+    the reference read path: LSMTree::get scans sstables newest-INDEX-first
+    and returns the first key match (lsm_tree.rs:692-696), so the highest
+    run index wins regardless of timestamp. Covers present, overwritten,
+    deleted and absent keys, plus the delete->get->KeyNotFound distinction
     (tests/db_server.rs:183-213 semantics)."""
     from dbeel_amd.engine import lookup
     from dbeel_amd.format import parse_run
@@ -226,13 +228,13 @@ def test_batched_lookup(engine):
                      seed=55)
     runs = [(bytes(d), bytes(i)) for d, i in runs]
 
-    # host model: newest (ts, run) wins
+    # host model: highest run index wins (reference read-path rule)
     model = {}
     for r, (d, i) in enumerate(runs):
         for e in parse_run(d, i):
             cur = model.get(e.key)
-            if cur is None or (e.timestamp, r) > cur[0]:
-                model[e.key] = ((e.timestamp, r), e.data)
+            if cur is None or r > cur[0]:
+                model[e.key] = (r, e.data)
 
     rng = np.random.default_rng(3)
     present = list(model.keys())
@@ -249,6 +251,26 @@ def test_batched_lookup(engine):
             assert g == exp, k.hex()
         else:
             assert g is None, k.hex()
+
+
+def test_lookup_newest_index_first(engine):
+    """The reference read path returns the match from the newest sstable
+    INDEX even when it carries an OLDER timestamp (set_with_timestamp via
+    replication can write one) — lsm_tree.rs:692-696 `.rev()` + first
+    match. The compaction winner rule (max timestamp) differs here; the
+    lookup must follow the read path, not the compaction rule."""
+    from dbeel_amd.engine import lookup
+    from dbeel_amd.format import Entry, build_run
+
+    k = b"duplicated-key!!"
+    run0 = build_run([Entry(k, b"newer-ts-old-run", 500)])
+    run1 = build_run([Entry(k, b"older-ts-new-run", 100)])
+    got = lookup([run0, run1], [k], device=0)
+    assert got == [b"older-ts-new-run"]
+    # tombstone in the newer run hides the older value (delete semantics)
+    run1t = build_run([Entry(k, b"", 100)])
+    got = lookup([run0, run1t], [k], device=0)
+    assert got == [b""]
 
 
 def test_resident_job_repeatable(engine):

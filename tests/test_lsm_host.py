@@ -114,6 +114,21 @@ def test_replay_multiple_actions_per_journal(tmp_path):
     assert open(f"{d}/{3:020d}.data", "rb").read() == b"B"
 
 
+def test_missing_input_files_are_io_errors(tmp_path):
+    """Host filesystem failures carry DBEEL_ERR_IO (7), distinct from
+    device failures (DBEEL_ERR_HIP) — an FFI caller mapping to the
+    reference's error.rs taxonomy must be able to tell them apart
+    (ADVICE r01). Reading absent run files fails before any device
+    access, so this tests on CPU."""
+    _skip_unless_built()
+    from dbeel_amd.engine import DbeelGpuError
+
+    with pytest.raises(DbeelGpuError) as ei:
+        lsm.compact(str(tmp_path), [0, 2], 1, keep_tombstones=False)
+    assert ei.value.code == 7  # IO
+    assert "IO" in str(ei.value)
+
+
 def test_bloom_contains_rejects_garbage():
     _skip_unless_built()
     from dbeel_amd.engine import DbeelGpuError
