@@ -253,6 +253,28 @@ def test_batched_lookup(engine):
             assert g is None, k.hex()
 
 
+def _adversarial_cases():
+    from adversarial import all_cases
+
+    return all_cases()
+
+
+@pytest.mark.parametrize("case", _adversarial_cases(),
+                         ids=lambda c: c[0])
+@pytest.mark.parametrize("keep", [True, False], ids=["keep", "drop"])
+def test_adversarial_cases_gpu(engine, case, keep):
+    """The adversarial seam cases (tests/adversarial.py — corank-window
+    straddles, 64-run identical keys, staged-prefix boundaries, long-key
+    blob-fallback ties, zero-padding families, i128 extremes) bit-exact
+    GPU vs oracle."""
+    name, runs = case
+    gd, gi, gn = engine.compact(runs, keep_tombstones=keep, device=0)
+    od, oi, on = oracle.compact(runs, keep_tombstones=keep)
+    assert gn == on, name
+    assert gi == oi, name
+    assert gd == od, name
+
+
 def test_lookup_newest_index_first(engine):
     """The reference read path returns the match from the newest sstable
     INDEX even when it carries an OLDER timestamp (set_with_timestamp via
