@@ -59,10 +59,10 @@ def make_workload(rank: int, scale: float, workload: str = "cfg3"):
     return [runs], cfg
 
 
-def rank_algorithmic_bytes(cfg, n_entries: int) -> int:
-    # k_rank minimum traffic: each entry's index record (16 B) + key +
-    # trailing timestamp (16 B) read once.
-    return n_entries * (16 + cfg["key_size"] + 16)
+def aux_record_bytes(key_size: int) -> int:
+    # aux tier (dbeel_gpu.hip AuxT): 16-B records for klen<=20, 32-B for
+    # klen<=36, 64-B beyond
+    return 16 if key_size <= 20 else (32 if key_size <= 36 else 64)
 
 
 def main():
@@ -191,12 +191,20 @@ def main():
     }
     dom = max(per_kernel, key=per_kernel.get)
     out_index_bytes = out_entries * 16
+    k = cfg["n_runs"]
+    aux = aux_record_bytes(cfg["key_size"])
     algo = {
-        # per-kernel algorithmic bytes
-        "prep": n_entries * (16 + cfg["key_size"] + 16 + 8),
-        "rank": rank_algorithmic_bytes(cfg, n_entries),
-        "scan": 3 * 8 * n_entries + 2 * 4 * n_entries,
-        "emit": 16 * n_entries + 24 * out_entries,
+        # per-kernel algorithmic bytes (r02 pipeline)
+        # prep: index rec + key bytes read; pfx (8) + aux written
+        "prep": n_entries * (16 + cfg["key_size"] + 8 + aux),
+        # corank: one pfx pass per pair side + cr written; rankreduce:
+        # index + pfx + aux + cr read, 16-B rrec written
+        "rank": n_entries * (2 * (k - 1) * 8 // max(k, 1) + 4 * (k - 1)
+                             + 16 + 8 + aux + 4 * (k - 1) + 16),
+        # fused survivor scan: rrec read once, SurvAgg written once
+        "scan": (16 + 16) * n_entries,
+        # emit: rrec + surv agg read, index rec + src_map written
+        "emit": 32 * n_entries + 24 * out_entries,
         "copy": 2 * out_bytes + 24 * out_entries,
     }
     dom_ms = per_kernel[dom]
@@ -206,10 +214,13 @@ def main():
     # against k_copy's known byte counts — see profiles/r01_traffic_cfg3.json)
     traffic = None
     pipe_traffic = None
-    tpath = os.path.join(os.path.dirname(os.path.abspath(__file__)),
-                         "profiles", "r01_traffic_cfg3.json")
-    if (args.scale == 1.0 and args.workload == "cfg3"
-            and os.path.exists(tpath)):
+    import glob as _glob
+
+    tcands = sorted(_glob.glob(os.path.join(
+        os.path.dirname(os.path.abspath(__file__)),
+        "profiles", "r*_traffic_cfg3.json")))
+    tpath = tcands[-1] if tcands else None
+    if args.scale == 1.0 and args.workload == "cfg3" and tpath:
         tj = json.load(open(tpath))
         kmap = tj.get("kernels", {})
         traffic = kmap.get(f"k_{dom}", {}).get("traffic_bytes")

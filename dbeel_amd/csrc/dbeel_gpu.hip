@@ -200,37 +200,52 @@ __device__ __forceinline__ uint64_t key_prefix(const uint8_t* key,
     return __builtin_bswap64(v);
 }
 
-/* Dense per-entry search record (one cache line for two): comparisons on
- * prefix-tied entries read THIS instead of the scattered multi-GiB data
- * blob. Keys longer than AUX_KEY_BYTES fall back to the blob only when two
- * such keys agree on their first AUX_KEY_BYTES bytes (true duplicates of
- * long keys). */
-#define AUX_KEY_BYTES 40
-struct Aux {
+/* Dense per-entry search record: the key's SUFFIX bytes (8..8+KB,
+ * zero-padded — the first 8 live in the pfx array, never duplicated here)
+ * plus the key length. Prefix-tied comparisons read THIS instead of the
+ * scattered multi-GiB data blob. Timestamps are NOT staged: the (ts, run)
+ * tie-break only fires on fully EQUAL keys (true cross-run duplicates),
+ * which read the two entries' trailing i128 from the blob — trading a
+ * rare scattered read for a dense 16-byte write saved on EVERY entry
+ * (cfg3: 1.0 GB of prep traffic). The tier (KB = 12/28/60 -> 16/32/64-B
+ * records) is picked per job from the max key_size in the uploaded
+ * indexes; longer keys stay CORRECT in any tier (blob fallback when two
+ * keys tie through 8+KB bytes), smaller tiers are purely a traffic
+ * optimization. */
+template <int KB> struct AuxT {
+    uint8_t key[KB]; /* key bytes 8..8+KB, zero-padded */
     uint32_t klen;
-    uint32_t rsv;
-    uint64_t ts_lo; /* timestamp i128 LE halves */
-    int64_t ts_hi;
-    uint8_t key[AUX_KEY_BYTES]; /* zero-padded */
 };
-static_assert(sizeof(Aux) == 64, "aux record must be 64B");
+static_assert(sizeof(AuxT<12>) == 16, "tier-0 aux must be 16B");
+static_assert(sizeof(AuxT<28>) == 32, "tier-1 aux must be 32B");
+static_assert(sizeof(AuxT<60>) == 64, "tier-2 aux must be 64B");
 
-/* Key-only compare of entries A=(rA,iA), B=(rB,iB) via aux records. */
-__device__ __forceinline__ int cmp_keys_aux(const RunsDesc& R,
-                                            const Aux* aux, int rA,
+/* Key-only compare of entries A=(rA,iA), B=(rB,iB) via suffix records.
+ * PRECONDITION: pfx(A) == pfx(B) (every caller compares pfx first), so
+ * the first min(8, klen) key bytes already agree and zero-padding in pfx
+ * is consistent: equality there with different klen means one key is a
+ * zero-extension of the other within 8 bytes, which the klen comparison
+ * below orders correctly (a strict prefix sorts first). */
+template <int KB>
+__device__ __forceinline__ int cmp_keys_sfx(const RunsDesc& R,
+                                            const AuxT<KB>* aux, int rA,
                                             uint64_t iA, int rB,
                                             uint64_t iB) {
-    const Aux* a = aux + R.entry_base[rA] + iA;
-    const Aux* b = aux + R.entry_base[rB] + iB;
+    const AuxT<KB>* a = aux + R.entry_base[rA] + iA;
+    const AuxT<KB>* b = aux + R.entry_base[rB] + iB;
     uint32_t la = a->klen, lb = b->klen;
     uint32_t n = la < lb ? la : lb;
-    if (n > AUX_KEY_BYTES) n = AUX_KEY_BYTES;
+    uint32_t ns = n > 8 ? n - 8 : 0; /* suffix bytes to compare */
+    if (ns > KB) ns = KB;
     #pragma unroll
-    for (uint32_t i = 0; i < AUX_KEY_BYTES; i += 8) {
-        if (i >= n) break;
-        uint64_t va = ld_u64(a->key + i), vb = ld_u64(b->key + i);
-        if (i + 8 > n) { /* mask the tail (pad bytes are zero anyway) */
-            uint64_t mask = (~0ull) >> (8 * (i + 8 - n));
+    for (uint32_t i = 0; i < KB; i += 8) {
+        if (i >= ns) break;
+        uint64_t va, vb;
+        __builtin_memcpy(&va, a->key + i, 8);
+        __builtin_memcpy(&vb, b->key + i, 8);
+        if (i + 8 > ns) { /* mask the tail; reading into klen is masked
+                             off (and pad bytes are zero anyway) */
+            uint64_t mask = (~0ull) >> (8 * (i + 8 - ns));
             va &= mask;
             vb &= mask;
         }
@@ -240,8 +255,8 @@ __device__ __forceinline__ int cmp_keys_aux(const RunsDesc& R,
             return va < vb ? -1 : 1;
         }
     }
-    if (la > AUX_KEY_BYTES && lb > AUX_KEY_BYTES) {
-        /* rare: long keys tied on the first AUX_KEY_BYTES bytes */
+    if (la > (uint32_t)(8 + KB) && lb > (uint32_t)(8 + KB)) {
+        /* rare: long keys tied through the staged 8+KB bytes */
         EView ea, eb;
         if (!load_entry(R, rA, iA, ea) || !load_entry(R, rB, iB, eb))
             return -1; /* corrupt input flagged elsewhere; value discarded */
@@ -250,26 +265,41 @@ __device__ __forceinline__ int cmp_keys_aux(const RunsDesc& R,
     return la < lb ? -1 : (la > lb ? 1 : 0);
 }
 
-/* Full-order compare (key, timestamp, run index — lsm_tree.rs:52-71) via
- * aux records. */
-__device__ __forceinline__ int cmp_aux_full(const RunsDesc& R,
-                                            const Aux* aux, int rA,
+/* (timestamp, run index) tie-break from the blob — only reached on fully
+ * equal keys (lsm_tree.rs:58-65 + mod.rs:75-81: ts i128 asc, then run
+ * index asc). */
+__device__ __forceinline__ int cmp_ts_run_blob(const RunsDesc& R, int rA,
+                                               uint64_t iA, int rB,
+                                               uint64_t iB) {
+    EView a, b;
+    if (!load_entry(R, rA, iA, a) || !load_entry(R, rB, iB, b))
+        return rA < rB ? -1 : 1; /* corrupt flagged elsewhere; stable */
+    uint64_t alo, blo;
+    int64_t ahi, bhi;
+    load_ts(a, alo, ahi);
+    load_ts(b, blo, bhi);
+    if (ahi != bhi) return ahi < bhi ? -1 : 1;
+    if (alo != blo) return alo < blo ? -1 : 1;
+    return rA < rB ? -1 : (rA > rB ? 1 : 0);
+}
+
+/* Full-order compare (key, timestamp, run index — lsm_tree.rs:52-71). */
+template <int KB>
+__device__ __forceinline__ int cmp_sfx_full(const RunsDesc& R,
+                                            const AuxT<KB>* aux, int rA,
                                             uint64_t iA, int rB,
                                             uint64_t iB) {
-    int c = cmp_keys_aux(R, aux, rA, iA, rB, iB);
+    int c = cmp_keys_sfx<KB>(R, aux, rA, iA, rB, iB);
     if (c) return c;
-    const Aux* a = aux + R.entry_base[rA] + iA;
-    const Aux* b = aux + R.entry_base[rB] + iB;
-    if (a->ts_hi != b->ts_hi) return a->ts_hi < b->ts_hi ? -1 : 1;
-    if (a->ts_lo != b->ts_lo) return a->ts_lo < b->ts_lo ? -1 : 1;
-    return rA < rB ? -1 : (rA > rB ? 1 : 0);
+    return cmp_ts_run_blob(R, rA, iA, rB, iB);
 }
 
 /* Validates every entry (bounds + bincode field cross-check,
  * read_next_entry lsm_tree.rs:1158-70) and extracts the dense key-prefix
  * and aux arrays the merge runs on. Run sortedness is checked in
  * k_rankreduce (on the dense pfx/aux arrays). */
-__global__ void k_prepare(RunsDesc R, uint64_t* pfx, Aux* aux,
+template <int KB>
+__global__ void k_prepare(RunsDesc R, uint64_t* pfx, AuxT<KB>* aux,
                           uint32_t* err) {
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
     for (uint64_t g = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
@@ -282,29 +312,29 @@ __global__ void k_prepare(RunsDesc R, uint64_t* pfx, Aux* aux,
         if (!load_entry(R, r, i, e)) {
             atomicOr(err, DERR_CORRUPT);
             pfx[g] = 0;
-            Aux z = {};
+            AuxT<KB> z = {};
             aux[g] = z;
             continue;
         }
         pfx[g] = key_prefix(e.key, e.klen);
-        Aux a;
+        AuxT<KB> a;
         a.klen = (uint32_t)e.klen;
-        a.rsv = 0;
-        load_ts(e, a.ts_lo, a.ts_hi);
-        uint32_t nk = e.klen < AUX_KEY_BYTES ? (uint32_t)e.klen
-                                             : AUX_KEY_BYTES;
-        /* 8-byte chunked key copy; reading up to 7 bytes past the key is
-         * safe (the 8-byte data_len field follows it inside the entry)
-         * and the over-read is masked off */
+        /* stage key SUFFIX bytes 8..8+KB, zero-padded. 8-byte chunked
+         * copy; reading up to 7 bytes past the key is safe (the 8-byte
+         * data_len field follows it inside the entry) and the over-read
+         * is masked off */
+        uint32_t nk = e.klen > 8 ? (uint32_t)e.klen - 8 : 0;
+        if (nk > KB) nk = KB;
         #pragma unroll
-        for (uint32_t j = 0; j < AUX_KEY_BYTES; j += 8) {
+        for (uint32_t j = 0; j < KB; j += 8) {
             uint64_t v = 0;
             if (j < nk) {
-                v = ld_u64(e.key + j);
+                v = ld_u64(e.key + 8 + j);
                 if (j + 8 > nk)
                     v &= (~0ull) >> (8 * (j + 8 - nk));
             }
-            __builtin_memcpy(a.key + j, &v, 8);
+            uint32_t nb = (KB - j) < 8 ? (KB - j) : 8;
+            __builtin_memcpy(a.key + j, &v, nb);
         }
         aux[g] = a;
         /* bincode field cross-check */
@@ -347,18 +377,11 @@ struct PairDesc {
     uint64_t chunk_base; /* exclusive prefix sum of per-pair windows */
 };
 
-/* strict full-order compare of a[ia] vs b[ib]; pfx values supplied by
- * the caller (staged in LDS or read globally) */
-__device__ __forceinline__ int cmp_tail_aux(const RunsDesc& R,
-                                            const Aux* aux, int ra,
-                                            uint64_t ia, int rb,
-                                            uint64_t ib) {
-    return cmp_aux_full(R, aux, ra, ia, rb, ib);
-}
-
+template <int KB>
 __global__ __launch_bounds__(CORANK_BLOCK) void k_corank(
-    RunsDesc R, const uint64_t* pfx, const Aux* aux, const PairDesc* pairs,
-    uint32_t n_pairs, uint64_t total_chunks, uint32_t* cr) {
+    RunsDesc R, const uint64_t* pfx, const AuxT<KB>* aux,
+    const PairDesc* pairs, uint32_t n_pairs, uint64_t total_chunks,
+    uint32_t* cr) {
     __shared__ uint64_t s_pfx[CORANK_BLOCK_POS + 2];
     __shared__ uint32_t s_cr[CORANK_BLOCK_POS]; /* staged crossranks:
         [0..lenA) for run a, [lenA..lenA+lenB) for run b — each block's
@@ -396,7 +419,8 @@ __global__ __launch_bounds__(CORANK_BLOCK) void k_corank(
                 uint64_t pa = gpa[mid], pb = gpb[diag - mid - 1];
                 int c = (pa != pb)
                             ? (pa < pb ? -1 : 1)
-                            : cmp_tail_aux(R, aux, a, mid, b, diag - mid - 1);
+                            : cmp_sfx_full<KB>(R, aux, a, mid, b,
+                                               diag - mid - 1);
                 if (c < 0)
                     slo = mid + 1;
                 else
@@ -431,8 +455,8 @@ __global__ __launch_bounds__(CORANK_BLOCK) void k_corank(
                 uint32_t mid = (slo + shi) >> 1;
                 uint64_t pa = sA[mid], pb = sB[d - mid - 1];
                 int c = (pa != pb) ? (pa < pb ? -1 : 1)
-                                   : cmp_tail_aux(R, aux, a, iaS + mid, b,
-                                                  ibS + d - mid - 1);
+                                   : cmp_sfx_full<KB>(R, aux, a, iaS + mid,
+                                                      b, ibS + d - mid - 1);
                 if (c < 0)
                     slo = mid + 1;
                 else
@@ -452,8 +476,8 @@ __global__ __launch_bounds__(CORANK_BLOCK) void k_corank(
                     uint64_t pa = sA[ja], pb = sB[jb];
                     take_a = (pa != pb)
                                  ? (pa < pb)
-                                 : (cmp_tail_aux(R, aux, a, iaS + ja, b,
-                                                 ibS + jb) < 0);
+                                 : (cmp_sfx_full<KB>(R, aux, a, iaS + ja, b,
+                                                     ibS + jb) < 0);
                 }
                 if (take_a) {
                     uint64_t gib = ibS + jb;
@@ -463,7 +487,8 @@ __global__ __launch_bounds__(CORANK_BLOCK) void k_corank(
                         uint64_t pb =
                             (jb < lenB) ? sB[jb] : gpb[gib];
                         if (pb == sA[ja] &&
-                            cmp_keys_aux(R, aux, a, iaS + ja, b, gib) == 0)
+                            cmp_keys_sfx<KB>(R, aux, a, iaS + ja, b,
+                                             gib) == 0)
                             v |= CR_LOSER;
                     }
                     s_cr[ja] = v;
@@ -475,7 +500,8 @@ __global__ __launch_bounds__(CORANK_BLOCK) void k_corank(
                         uint64_t pa =
                             (ja < lenA) ? sA[ja] : gpa[gia];
                         if (pa == sB[jb] &&
-                            cmp_keys_aux(R, aux, b, ibS + jb, a, gia) == 0)
+                            cmp_keys_sfx<KB>(R, aux, b, ibS + jb, a,
+                                             gia) == 0)
                             v |= CR_LOSER;
                     }
                     s_cr[lenA + jb] = v;
@@ -501,25 +527,25 @@ __global__ __launch_bounds__(CORANK_BLOCK) void k_corank(
     }
 }
 
-/* Rank-indexed scratch record: one 32-B scattered write per entry
- * (vs three separate arrays = 3 scattered lines). size == 0 marks a
- * dropped entry; the survivor scans read it through strided transform
- * iterators. */
+/* Rank-indexed scratch record: one 16-B scattered write per entry.
+ * src packs keep flag (bit 63) | run (bits 48..62) | data offset (48).
+ * The survivor scan reads it through a transform iterator. */
 struct RankRec {
-    uint64_t src;      /* run (high 16 bits) | data offset (low 48) */
-    uint32_t key_size; /* 8 + key_len (index record field)          */
+    uint64_t src;
+    uint32_t key_size; /* 8 + key_len (index record field) */
     uint32_t full_size;
-    uint64_t size;     /* full_size if kept, else 0                 */
-    uint64_t pad;
 };
-static_assert(sizeof(RankRec) == 32, "rank record must be 32B");
+static_assert(sizeof(RankRec) == 16, "rank record must be 16B");
+#define RR_KEEP (1ull << 63)
+#define RR_OFF_MASK 0xFFFFFFFFFFFFull
 
 /* Reduce per-pair crossranks to the global rank, apply the winner /
  * tombstone rules, and emit the rank-indexed scratch records. Also checks
  * each run is strictly sorted by key (flush invariant,
  * lsm_tree.rs:925-946) on the dense pfx/aux arrays. */
+template <int KB>
 __global__ void k_rankreduce(RunsDesc R, const uint64_t* pfx,
-                             const Aux* aux, const uint32_t* cr,
+                             const AuxT<KB>* aux, const uint32_t* cr,
                              RankRec* rrec, int keep_tombstones,
                              uint32_t* err) {
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
@@ -532,14 +558,15 @@ __global__ void k_rankreduce(RunsDesc R, const uint64_t* pfx,
         if (!load_entry(R, r, i, e)) {
             /* k_prepare has flagged this input; keep memory safe and park
              * the entry at its local slot (results will be discarded) */
-            RankRec z = {0, 8, 32, 0, 0};
+            RankRec z = {0, 8, 32};
             rrec[g] = z;
             continue;
         }
         if (i + 1 < R.count[r]) {
             uint64_t p0 = pfx[g], p1 = pfx[g + 1];
             if (p0 > p1 ||
-                (p0 == p1 && cmp_keys_aux(R, aux, r, i, r, i + 1) >= 0))
+                (p0 == p1 &&
+                 cmp_keys_sfx<KB>(R, aux, r, i, r, i + 1) >= 0))
                 atomicOr(err, DERR_UNSORTED);
         }
 
@@ -553,47 +580,54 @@ __global__ void k_rankreduce(RunsDesc R, const uint64_t* pfx,
         uint64_t dlen = (uint64_t)e.full_size - 32 - e.klen;
         bool keep = winner && (keep_tombstones || dlen != 0);
         RankRec m;
-        m.src = ((uint64_t)r << 48) | e.off;
+        m.src = ((uint64_t)r << 48) | e.off | (keep ? RR_KEEP : 0);
         m.key_size = e.key_size;
         m.full_size = e.full_size;
-        m.size = keep ? e.full_size : 0;
-        m.pad = 0;
         rrec[rank] = m;
     }
 }
 
+/* Survivor aggregate for the single fused scan: output byte offset and
+ * survivor position in one pass over rrec (one 16-B read + one 16-B
+ * write per entry, replacing the former two 32-B-read passes). */
+struct SurvAgg {
+    uint64_t bytes;
+    uint64_t cnt;
+};
+struct SurvAggPlus {
+    __device__ SurvAgg operator()(const SurvAgg& a, const SurvAgg& b) const {
+        return {a.bytes + b.bytes, a.cnt + b.cnt};
+    }
+};
+struct RankRecToAgg {
+    __device__ SurvAgg operator()(const RankRec& r) const {
+        uint64_t k = (r.src & RR_KEEP) ? 1u : 0u;
+        return {k ? (uint64_t)r.full_size : 0ull, k};
+    }
+};
+
 /* Output index records are the input format: offset u64 | key_size u32 |
  * full_size u32 (entry_writer.rs:79-87, offsets recomputed from 0).
  * src_map gets the ABSOLUTE device address of each survivor's bytes. */
-__global__ void k_emit(RunsDesc R, const RankRec* rrec,
-                       const uint64_t* dst_off, const uint32_t* pos,
+__global__ void k_emit(RunsDesc R, const RankRec* rrec, const SurvAgg* surv,
                        uint64_t total, uint8_t* out_index,
                        uint64_t* src_map) {
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
     for (uint64_t g = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
          g < total; g += stride) {
         RankRec m = rrec[g];
-        if (!m.size) continue;
-        uint32_t p = pos[g];
-        uint8_t* rec = out_index + (uint64_t)p * 16;
-        uint64_t off = dst_off[g];
+        if (!(m.src & RR_KEEP)) continue;
+        SurvAgg s = surv[g];
+        uint64_t p = s.cnt;
+        uint8_t* rec = out_index + p * 16;
+        uint64_t off = s.bytes;
         __builtin_memcpy(rec, &off, 8);
         __builtin_memcpy(rec + 8, &m.key_size, 4);
         __builtin_memcpy(rec + 12, &m.full_size, 4);
-        src_map[p] =
-            (uint64_t)(R.data[m.src >> 48] + (m.src & 0xFFFFFFFFFFFFull));
+        src_map[p] = (uint64_t)(R.data[(m.src >> 48) & 0x7FFF] +
+                                (m.src & RR_OFF_MASK));
     }
 }
-
-/* transform-iterator functors for the survivor scans */
-struct RankRecSize {
-    __device__ uint64_t operator()(const RankRec& r) const { return r.size; }
-};
-struct RankRecFlag {
-    __device__ uint32_t operator()(const RankRec& r) const {
-        return r.size ? 1u : 0u;
-    }
-};
 
 /* Balanced verbatim copy: a 16-KiB destination window per 256-thread
  * block, 4 x 16-B granules per thread (independent loads/stores for ILP;
@@ -698,76 +732,92 @@ __global__ __launch_bounds__(BLK) void k_copy(
         }
         __syncthreads();
 
-        #pragma unroll
-        for (int q = 0; q < GRAN / BLK; q++) {
-            uint32_t gl = q * BLK + threadIdx.x; /* window-local */
-            uint64_t gpos = wstart + (uint64_t)gl * 16;
-            if (gpos >= total_bytes) break;
-            uint32_t j = s_gid[gl];
-            uint64_t e_end = s_off[j + 1];
-            const uint8_t* src = (const uint8_t*)s_src[j] +
-                                 (gpos - s_off[j]);
-            uint32_t nbytes =
-                (uint32_t)((total_bytes - gpos) < 16 ? (total_bytes - gpos)
-                                                     : 16);
-            uint8_t* dst = out_data + gpos;
-            if (gpos + nbytes <= e_end) {
-                if (nbytes == 16) {
-                    typedef unsigned int v4u
-                        __attribute__((ext_vector_type(4)));
-                    v4u v;
-                    __builtin_memcpy(&v, src, 16);
-                    /* streamed once, never re-read: keep L2 for sources */
-                    __builtin_nontemporal_store(
-                        v, reinterpret_cast<v4u*>(dst));
+        typedef unsigned int v4u __attribute__((ext_vector_type(4)));
+        if (wstart + WIN <= total_bytes) {
+            /* full interior window: every granule is a whole 16 B. Two
+             * phases — gather all GRAN/BLK granule values into registers
+             * first (independent load chains, keeps several loads in
+             * flight per wave), then store. The single-phase form left
+             * most waves parked on one load at a time (93% parked on the
+             * 4 KiB-value shape, r01 profile). */
+            v4u vv[GRAN / BLK];
+            #pragma unroll
+            for (int q = 0; q < GRAN / BLK; q++) {
+                uint32_t gl = q * BLK + threadIdx.x; /* window-local */
+                uint64_t gpos = wstart + (uint64_t)gl * 16;
+                uint32_t j = s_gid[gl];
+                uint64_t e_end = s_off[j + 1];
+                const uint8_t* src = (const uint8_t*)s_src[j] +
+                                     (gpos - s_off[j]);
+                if (gpos + 16 <= e_end) {
+                    __builtin_memcpy(&vv[q], src, 16);
                 } else {
-                    for (uint32_t b = 0; b < nbytes; b++) dst[b] = src[b];
-                }
-            } else if (nbytes == 16) {
-                /* one entry boundary inside the granule: 16-B blend of the
-                 * entry tail and the next survivor's head. Byte loops here
-                 * cost a whole wave ~3k serial cycles, and odd entry sizes
-                 * put one straddle in nearly every wave (measured 2.4x on
-                 * odd-size values). Reading past either entry stays inside
-                 * the padded input slab. */
-                uint32_t c1 = (uint32_t)(e_end - gpos); /* 1..15 */
-                const uint8_t* src2 = (const uint8_t*)s_src[j + 1];
-                uint64_t l0, h0, l2, h2;
-                __builtin_memcpy(&l0, src, 8);
-                __builtin_memcpy(&h0, src + 8, 8);
-                __builtin_memcpy(&l2, src2, 8);
-                __builtin_memcpy(&h2, src2 + 8, 8);
-                uint64_t lo, hi;
-                if (c1 < 8) {
-                    uint32_t s = 8 * c1;
-                    uint64_t m = (~0ull) >> (64 - s);
-                    lo = (l0 & m) | (l2 << s);
-                    hi = (h2 << s) | (l2 >> (64 - s));
-                } else {
-                    uint32_t c = c1 - 8; /* 0..7 */
-                    lo = l0;
-                    if (c == 0) {
-                        hi = l2;
+                    /* one entry boundary inside the granule: 16-B blend
+                     * of the entry tail and the next survivor's head.
+                     * Byte loops here cost a whole wave ~3k serial
+                     * cycles, and odd entry sizes put one straddle in
+                     * nearly every wave (measured 2.4x on odd-size
+                     * values). Reading past either entry stays inside
+                     * the padded input slab. */
+                    uint32_t c1 = (uint32_t)(e_end - gpos); /* 1..15 */
+                    const uint8_t* src2 = (const uint8_t*)s_src[j + 1];
+                    uint64_t l0, h0, l2, h2;
+                    __builtin_memcpy(&l0, src, 8);
+                    __builtin_memcpy(&h0, src + 8, 8);
+                    __builtin_memcpy(&l2, src2, 8);
+                    __builtin_memcpy(&h2, src2 + 8, 8);
+                    uint64_t lo, hi;
+                    if (c1 < 8) {
+                        uint32_t sh = 8 * c1;
+                        uint64_t m = (~0ull) >> (64 - sh);
+                        lo = (l0 & m) | (l2 << sh);
+                        hi = (h2 << sh) | (l2 >> (64 - sh));
                     } else {
-                        uint64_t m = (~0ull) >> (64 - 8 * c);
-                        hi = (h0 & m) | (l2 << (8 * c));
+                        uint32_t c = c1 - 8; /* 0..7 */
+                        lo = l0;
+                        if (c == 0) {
+                            hi = l2;
+                        } else {
+                            uint64_t m = (~0ull) >> (64 - 8 * c);
+                            hi = (h0 & m) | (l2 << (8 * c));
+                        }
                     }
+                    __builtin_memcpy(&vv[q], &lo, 8);
+                    __builtin_memcpy(
+                        reinterpret_cast<uint8_t*>(&vv[q]) + 8, &hi, 8);
                 }
-                typedef unsigned int v4u
-                    __attribute__((ext_vector_type(4)));
-                v4u v;
-                __builtin_memcpy(&v, &lo, 8);
-                __builtin_memcpy(reinterpret_cast<uint8_t*>(&v) + 8, &hi, 8);
-                __builtin_nontemporal_store(v,
-                                            reinterpret_cast<v4u*>(dst));
-            } else {
-                /* output tail granule (< 16 B, at most one per job) */
-                uint32_t c1 = (uint32_t)(e_end - gpos);
-                if (c1 > nbytes) c1 = nbytes;
+            }
+            /* streamed once, never re-read: keep L2 for sources */
+            #pragma unroll
+            for (int q = 0; q < GRAN / BLK; q++) {
+                uint32_t gl = q * BLK + threadIdx.x;
+                __builtin_nontemporal_store(
+                    vv[q], reinterpret_cast<v4u*>(out_data + wstart +
+                                                  (uint64_t)gl * 16));
+            }
+        } else {
+            /* the job's final (partial) window */
+            #pragma unroll
+            for (int q = 0; q < GRAN / BLK; q++) {
+                uint32_t gl = q * BLK + threadIdx.x;
+                uint64_t gpos = wstart + (uint64_t)gl * 16;
+                if (gpos >= total_bytes) break;
+                uint32_t j = s_gid[gl];
+                uint64_t e_end = s_off[j + 1];
+                const uint8_t* src = (const uint8_t*)s_src[j] +
+                                     (gpos - s_off[j]);
+                uint32_t nbytes = (uint32_t)(
+                    (total_bytes - gpos) < 16 ? (total_bytes - gpos) : 16);
+                uint8_t* dst = out_data + gpos;
+                uint32_t c1 = (uint32_t)(gpos + nbytes <= e_end
+                                             ? nbytes
+                                             : e_end - gpos);
                 for (uint32_t b = 0; b < c1; b++) dst[b] = src[b];
-                const uint8_t* src2 = (const uint8_t*)s_src[j + 1];
-                for (uint32_t b = c1; b < nbytes; b++)
-                    dst[b] = src2[b - c1];
+                if (c1 < nbytes) {
+                    const uint8_t* src2 = (const uint8_t*)s_src[j + 1];
+                    for (uint32_t b = c1; b < nbytes; b++)
+                        dst[b] = src2[b - c1];
+                }
             }
         }
         __syncthreads();
@@ -992,6 +1042,26 @@ done:
 /* Host side                                                          */
 /* ------------------------------------------------------------------ */
 
+/* Max key_size across all uploaded index slabs — picks the aux tier. */
+__global__ void k_maxks(RunsDesc R, uint32_t* out) {
+    __shared__ uint32_t smax;
+    if (threadIdx.x == 0) smax = 0;
+    __syncthreads();
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    uint32_t m = 0;
+    for (uint64_t g = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         g < R.total; g += stride) {
+        int r = 0;
+        while (r + 1 < R.n_runs && g >= R.entry_base[r + 1]) r++;
+        uint64_t i = g - R.entry_base[r];
+        uint32_t ks = ld_u32(R.index[r] + i * 16 + 8);
+        if (ks > m) m = ks;
+    }
+    atomicMax(&smax, m);
+    __syncthreads();
+    if (threadIdx.x == 0) atomicMax(out, smax);
+}
+
 struct dbeel_gpu_job {
     int device = -1;
     hipStream_t stream = nullptr;
@@ -999,8 +1069,8 @@ struct dbeel_gpu_job {
     RunsDesc desc{};
     uint8_t* d_input = nullptr; /* one slab: all run data+index            */
     RankRec* d_rank = nullptr;
-    uint64_t* d_dstoff = nullptr;
-    uint32_t* d_pos = nullptr;
+    SurvAgg* d_surv = nullptr;  /* fused survivor scan output              */
+    int aux_kind = 2;           /* 0: 16-B, 1: 32-B, 2: 64-B aux records   */
     uint8_t* d_outindex = nullptr;
     uint64_t* d_srcmap = nullptr;
     uint8_t* d_outdata = nullptr;
@@ -1124,32 +1194,25 @@ extern "C" int dbeel_gpu_job_create(const dbeel_run_view* runs, size_t n_runs,
     JOB_CHECK(hipMalloc(&job->d_input, (input_bytes ? input_bytes : 16) + 32));
     uint64_t n = total ? total : 1;
     JOB_CHECK(hipMalloc(&job->d_rank, n * sizeof(RankRec)));
-    JOB_CHECK(hipMalloc(&job->d_dstoff, n * sizeof(uint64_t)));
-    JOB_CHECK(hipMalloc(&job->d_pos, n * sizeof(uint32_t)));
+    JOB_CHECK(hipMalloc(&job->d_surv, n * sizeof(SurvAgg)));
     JOB_CHECK(hipMalloc(&job->d_outindex, n * 16));
     JOB_CHECK(hipMalloc(&job->d_srcmap, n * sizeof(uint64_t)));
     JOB_CHECK(hipMalloc(&job->d_outdata, total_data ? total_data : 16));
     JOB_CHECK(hipMalloc(&job->d_err, 2 * sizeof(uint32_t)));
     JOB_CHECK(hipMalloc(&job->d_pfx, n * sizeof(uint64_t)));
-    JOB_CHECK(hipMalloc(&job->d_aux, n * 64));
     JOB_CHECK(hipMalloc(&job->d_cr,
                         (n_runs > 1 ? (n_runs - 1) * n : 1) * 4));
     /* sized for the smallest copy-window variant (8 KiB) */
     JOB_CHECK(hipMalloc(&job->d_winp0,
                         (total_data / 8192 + 2) * sizeof(uint32_t)));
 
-    size_t t1 = 0, t2 = 0;
+    size_t t1 = 0;
     rocprim::exclusive_scan(nullptr, t1,
                             rocprim::make_transform_iterator(job->d_rank,
-                                                             RankRecSize{}),
-                            job->d_dstoff, (uint64_t)0, n,
-                            rocprim::plus<uint64_t>(), job->stream);
-    rocprim::exclusive_scan(nullptr, t2,
-                            rocprim::make_transform_iterator(job->d_rank,
-                                                             RankRecFlag{}),
-                            job->d_pos, (uint32_t)0, n,
-                            rocprim::plus<uint32_t>(), job->stream);
-    job->scantmp_bytes = t1 > t2 ? t1 : t2;
+                                                             RankRecToAgg{}),
+                            job->d_surv, SurvAgg{0, 0}, n, SurvAggPlus{},
+                            job->stream);
+    job->scantmp_bytes = t1;
     JOB_CHECK(hipMalloc(&job->d_scantmp, job->scantmp_bytes));
 
     /* Upload: one slab; record per-run device pointers. */
@@ -1223,6 +1286,26 @@ extern "C" int dbeel_gpu_job_create(const dbeel_run_view* runs, size_t n_runs,
     JOB_CHECK(hipEventElapsedTime(&ms, job->ev[0], job->ev[1]));
     job->h2d_ms = ms;
 
+    /* pick the aux tier from the max key_size (index slabs are resident
+     * now); the tier is a traffic optimization only — every tier is
+     * correct for every key length (blob fallback on deep ties) */
+    if (total) {
+        JOB_CHECK(hipMemsetAsync(job->d_err, 0, 8, job->stream));
+        hipLaunchKernelGGL(k_maxks, dim3(pick_grid(total, 256)), dim3(256),
+                           0, job->stream, job->desc, job->d_err + 1);
+        uint32_t maxks = 0;
+        JOB_CHECK(hipMemcpyAsync(&maxks, job->d_err + 1, 4,
+                                 hipMemcpyDeviceToHost, job->stream));
+        JOB_CHECK(hipStreamSynchronize(job->stream));
+        uint32_t maxklen = maxks > 8 ? maxks - 8 : 0;
+        job->aux_kind = maxklen <= 20 ? 0 : (maxklen <= 36 ? 1 : 2);
+    } else {
+        job->aux_kind = 0;
+    }
+    uint64_t aux_sz = job->aux_kind == 0 ? 16 : (job->aux_kind == 1 ? 32
+                                                                    : 64);
+    JOB_CHECK(hipMalloc(&job->d_aux, n * aux_sz));
+
     *out_job = job;
     return DBEEL_OK;
 #undef JOB_CHECK
@@ -1233,8 +1316,7 @@ extern "C" void dbeel_gpu_job_destroy(dbeel_gpu_job* job) {
     if (job->device >= 0) hipSetDevice(job->device);
     hipFree(job->d_input);
     hipFree(job->d_rank);
-    hipFree(job->d_dstoff);
-    hipFree(job->d_pos);
+    hipFree(job->d_surv);
     hipFree(job->d_outindex);
     hipFree(job->d_srcmap);
     hipFree(job->d_outdata);
@@ -1278,57 +1360,59 @@ extern "C" int dbeel_gpu_job_run(dbeel_gpu_job* job, int keep_tombstones,
     HIP_CHECK(hipEventRecord(job->ev[0], s));
     if (n) {
         uint32_t grid = pick_grid(n, 256);
-        hipLaunchKernelGGL(k_prepare, dim3(grid), dim3(256), 0, s, job->desc,
-                           job->d_pfx, (Aux*)job->d_aux, job->d_err);
-    }
-    HIP_CHECK(hipEventRecord(job->ev[6], s));
-    if (n) {
-        uint32_t grid = pick_grid(n, 256);
-        if (job->n_pairs) {
-            uint64_t cgrid = job->total_chunks;
-            if (cgrid > 16384) cgrid = 16384;
-            hipLaunchKernelGGL(k_corank, dim3((uint32_t)cgrid),
-                               dim3(CORANK_BLOCK), 0, s, job->desc,
-                               job->d_pfx, (const Aux*)job->d_aux,
-                               (const PairDesc*)job->d_pairs, job->n_pairs,
-                               job->total_chunks, job->d_cr);
+        uint64_t cgrid = job->total_chunks;
+        if (cgrid > 16384) cgrid = 16384;
+        switch (job->aux_kind) {
+#define STAGE1(KB)                                                          \
+    case (KB == 12 ? 0 : (KB == 28 ? 1 : 2)):                               \
+        hipLaunchKernelGGL(k_prepare<KB>, dim3(grid), dim3(256), 0, s,      \
+                           job->desc, job->d_pfx, (AuxT<KB>*)job->d_aux,    \
+                           job->d_err);                                     \
+        hipEventRecord(job->ev[6], s);                                      \
+        if (job->n_pairs)                                                   \
+            hipLaunchKernelGGL(k_corank<KB>, dim3((uint32_t)cgrid),         \
+                               dim3(CORANK_BLOCK), 0, s, job->desc,         \
+                               job->d_pfx, (const AuxT<KB>*)job->d_aux,     \
+                               (const PairDesc*)job->d_pairs, job->n_pairs, \
+                               job->total_chunks, job->d_cr);               \
+        hipLaunchKernelGGL(k_rankreduce<KB>, dim3(grid), dim3(256), 0, s,   \
+                           job->desc, job->d_pfx,                           \
+                           (const AuxT<KB>*)job->d_aux, job->d_cr,          \
+                           job->d_rank, keep_tombstones, job->d_err);       \
+        break;
+            STAGE1(12)
+            STAGE1(28)
+            STAGE1(60)
+#undef STAGE1
         }
-        hipLaunchKernelGGL(k_rankreduce, dim3(grid), dim3(256), 0, s,
-                           job->desc, job->d_pfx, (const Aux*)job->d_aux,
-                           job->d_cr, job->d_rank, keep_tombstones,
-                           job->d_err);
+    } else {
+        HIP_CHECK(hipEventRecord(job->ev[6], s));
     }
     HIP_CHECK(hipEventRecord(job->ev[1], s));
     if (n) {
         size_t tmp = job->scantmp_bytes;
         (void)rocprim::exclusive_scan(
             job->d_scantmp, tmp,
-            rocprim::make_transform_iterator(job->d_rank, RankRecSize{}),
-            job->d_dstoff, (uint64_t)0, n, rocprim::plus<uint64_t>(), s);
-        tmp = job->scantmp_bytes;
-        (void)rocprim::exclusive_scan(
-            job->d_scantmp, tmp,
-            rocprim::make_transform_iterator(job->d_rank, RankRecFlag{}),
-            job->d_pos, (uint32_t)0, n, rocprim::plus<uint32_t>(), s);
+            rocprim::make_transform_iterator(job->d_rank, RankRecToAgg{}),
+            job->d_surv, SurvAgg{0, 0}, n, SurvAggPlus{}, s);
     }
     HIP_CHECK(hipEventRecord(job->ev[2], s));
     if (n) {
         uint32_t grid = pick_grid(n, 256);
         hipLaunchKernelGGL(k_emit, dim3(grid), dim3(256), 0, s, job->desc,
-                           job->d_rank, job->d_dstoff, job->d_pos, n,
-                           job->d_outindex, job->d_srcmap);
+                           job->d_rank, job->d_surv, n, job->d_outindex,
+                           job->d_srcmap);
     }
     HIP_CHECK(hipEventRecord(job->ev[3], s));
 
     /* Need totals on host to size/launch the copy; one small sync. */
-    uint64_t last_size = 0, last_off = 0;
-    uint32_t last_pos = 0, err = 0;
+    RankRec last_rec = {};
+    SurvAgg last_agg = {};
+    uint32_t err = 0;
     if (n) {
-        HIP_CHECK(hipMemcpyAsync(&last_size, &job->d_rank[n - 1].size, 8,
+        HIP_CHECK(hipMemcpyAsync(&last_rec, job->d_rank + (n - 1), 16,
                                  hipMemcpyDeviceToHost, s));
-        HIP_CHECK(hipMemcpyAsync(&last_off, job->d_dstoff + (n - 1), 8,
-                                 hipMemcpyDeviceToHost, s));
-        HIP_CHECK(hipMemcpyAsync(&last_pos, job->d_pos + (n - 1), 4,
+        HIP_CHECK(hipMemcpyAsync(&last_agg, job->d_surv + (n - 1), 16,
                                  hipMemcpyDeviceToHost, s));
     }
     HIP_CHECK(hipMemcpyAsync(&err, job->d_err, 4, hipMemcpyDeviceToHost, s));
@@ -1340,8 +1424,10 @@ extern "C" int dbeel_gpu_job_run(dbeel_gpu_job* job, int keep_tombstones,
                     : "corrupt entry/index record");
         return DBEEL_ERR_CORRUPT;
     }
-    uint64_t total_out = last_off + last_size;
-    uint64_t n_surv = (uint64_t)last_pos + (last_size ? 1 : 0);
+    uint64_t last_kept = (n && (last_rec.src & RR_KEEP)) ? 1 : 0;
+    uint64_t total_out =
+        last_agg.bytes + (last_kept ? last_rec.full_size : 0);
+    uint64_t n_surv = last_agg.cnt + last_kept;
 
     HIP_CHECK(hipEventRecord(job->ev[4], s));
     if (total_out) {
