@@ -52,6 +52,12 @@ def make_workload(rank: int, scale: float, workload: str = "cfg3"):
             for j in range(8)
         ]
         return jobs, cfg
+    if workload == "cfg2":
+        cfg = dict(CONFIGS["cfg2"])
+        if scale != 1.0:
+            cfg["entries_per_run"] = max(64, int(cfg["entries_per_run"] * scale))
+        runs = make_runs(seed=0xDBEE1 + 7919 * rank, **cfg)
+        return [runs], cfg
     cfg = dict(CONFIGS["cfg3"])
     if scale != 1.0:
         cfg["entries_per_run"] = max(64, int(cfg["entries_per_run"] * scale))
@@ -72,11 +78,14 @@ def main():
     ap.add_argument("--warmup", type=int, default=3)
     ap.add_argument("--scale", type=float, default=1.0,
                     help="scale entries_per_run (1.0 = full config 3)")
-    ap.add_argument("--workload", choices=["cfg3", "cfg4", "cfg5"],
+    ap.add_argument("--workload", choices=["cfg2", "cfg3", "cfg4", "cfg5"],
                     default="cfg3",
-                    help="cfg3 = 8-run x 1 GiB metric config; cfg4 = 8 "
+                    help="cfg2 = 4-run x 1M x 304 B (BASELINE configs[1]); "
+                         "cfg3 = 8-run x 1 GiB metric config; cfg4 = 8 "
                          "independent 4-run x 256 MiB jobs per GPU; cfg5 = "
                          "16-run var-length msgpack keys + 4 KiB values")
+    ap.add_argument("--skip-streamed", action="store_true",
+                    help="skip the end-to-end streamed-ingest measurement")
     ap.add_argument("--keep-tombstones", action="store_true")
     ap.add_argument("--cpu-baseline-scale", type=float, default=1.0,
                     help="fraction of the workload timed on the host "
@@ -201,10 +210,10 @@ def main():
         # index + pfx + aux + cr read, 16-B rrec written
         "rank": n_entries * (2 * (k - 1) * 8 // max(k, 1) + 4 * (k - 1)
                              + 16 + 8 + aux + 4 * (k - 1) + 16),
-        # fused survivor scan: rrec read once, SurvAgg written once
-        "scan": (16 + 16) * n_entries,
-        # emit: rrec + surv agg read, index rec + src_map written
-        "emit": 32 * n_entries + 24 * out_entries,
+        # two survivor scans: 16-B rrec read per pass, offsets written
+        "scan": (2 * 16 + 8 + 4) * n_entries,
+        # emit: rrec + offsets read, index rec + src_map written
+        "emit": 28 * n_entries + 24 * out_entries,
         "copy": 2 * out_bytes + 24 * out_entries,
     }
     dom_ms = per_kernel[dom]
@@ -245,6 +254,52 @@ def main():
         "frac": round(pipe_gbps / HBM_PEAK_GBPS, 4),
         "traffic": pipe_traffic,
     }
+
+    # End-to-end streamed ingest (north_star: pinned host DRAM ->
+    # hipMemcpyAsync chunks overlapped with the prepare kernel). Measured
+    # beside the resident metric ("end-to-end MB/s incl.
+    # PCIe streaming is also reported"); never `value`.
+    end_to_end = None
+    if (rank == 0 and n_gpus == 1 and not args.skip_streamed
+            and len(jobs) == 1):
+        from dbeel_amd.engine import pin_host, unpin_host
+
+        bufs = [a for d, i in job_runs[0] for a in (d, i)]
+        try:
+            for a in bufs:
+                pin_host(a)
+            pinned = True
+        except Exception:
+            pinned = False  # fall back to pageable streaming
+        try:
+            jobs[0].ingest(job_runs[0])  # warmup
+            jobs[0].run(keep)
+            e2e_steps = 3
+            ist_acc = None
+            t0 = time.perf_counter()
+            for _ in range(e2e_steps):
+                ist = jobs[0].ingest(job_runs[0])
+                jobs[0].run(keep)
+                ist_acc = ist if ist_acc is None else {
+                    k: ist_acc[k] + ist[k] for k in ist
+                }
+            e2e_s = time.perf_counter() - t0
+            end_to_end = {
+                "MBps": round((input_bytes / 1e6) / (e2e_s / e2e_steps), 1),
+                "ingest_ms": round(ist_acc["ingest_ms"] / e2e_steps, 3),
+                "h2d_copy_ms": round(ist_acc["copy_ms"] / e2e_steps, 3),
+                "prep_hidden_in_ingest": True,
+                "chunks": int(ist_acc["chunks"] // e2e_steps),
+                "pinned": pinned,
+                "steps": e2e_steps,
+            }
+        finally:
+            if pinned:
+                for a in bufs:
+                    try:
+                        unpin_host(a)
+                    except Exception:
+                        pass
 
     cpu_baseline = None
     if rank == 0 and n_gpus == 1 and not args.skip_cpu_baseline:
@@ -316,6 +371,7 @@ def main():
             "out_entries": out_entries,
             "roofline": roofline,
             "roofline_pipeline": roofline_pipeline,
+            "end_to_end": end_to_end,
             "cpu_baseline": cpu_baseline,
         }
         print(json.dumps(line))

@@ -45,6 +45,7 @@
 #include <cstring>
 #include <cstdlib>
 #include <new>
+#include <vector>
 
 #include <hip/hip_runtime.h>
 #include <rocprim/rocprim.hpp>
@@ -299,11 +300,11 @@ __device__ __forceinline__ int cmp_sfx_full(const RunsDesc& R,
  * and aux arrays the merge runs on. Run sortedness is checked in
  * k_rankreduce (on the dense pfx/aux arrays). */
 template <int KB>
-__global__ void k_prepare(RunsDesc R, uint64_t* pfx, AuxT<KB>* aux,
-                          uint32_t* err) {
+__global__ void k_prepare(RunsDesc R, uint64_t g0, uint64_t g1,
+                          uint64_t* pfx, AuxT<KB>* aux, uint32_t* err) {
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
-    for (uint64_t g = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
-         g < R.total; g += stride) {
+    for (uint64_t g = g0 + (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         g < g1; g += stride) {
         /* locate run (<= 64 runs: linear scan on cached desc) */
         int r = 0;
         while (r + 1 < R.n_runs && g >= R.entry_base[r + 1]) r++;
@@ -341,6 +342,14 @@ __global__ void k_prepare(RunsDesc R, uint64_t* pfx, AuxT<KB>* aux,
         if (ld_u64(e.raw) != e.klen ||
             ld_u64(e.raw + 8 + e.klen) != (uint64_t)e.full_size - 32 - e.klen)
             atomicOr(err, DERR_CORRUPT);
+        /* entries within a run must not overlap and offsets must be
+         * monotone (EntryWriter appends — entry_writer.rs:71-98; the
+         * streamed-ingest chunking relies on this invariant) */
+        if (i + 1 < R.count[r]) {
+            uint64_t noff = ld_u64(R.index[r] + (i + 1) * 16);
+            if (e.off + e.full_size > noff)
+                atomicOr(err, DERR_CORRUPT);
+        }
     }
 }
 
@@ -587,29 +596,25 @@ __global__ void k_rankreduce(RunsDesc R, const uint64_t* pfx,
     }
 }
 
-/* Survivor aggregate for the single fused scan: output byte offset and
- * survivor position in one pass over rrec (one 16-B read + one 16-B
- * write per entry, replacing the former two 32-B-read passes). */
-struct SurvAgg {
-    uint64_t bytes;
-    uint64_t cnt;
-};
-struct SurvAggPlus {
-    __device__ SurvAgg operator()(const SurvAgg& a, const SurvAgg& b) const {
-        return {a.bytes + b.bytes, a.cnt + b.cnt};
+/* transform-iterator functors for the two survivor scans (plain
+ * arithmetic types + rocprim::plus keep rocPRIM on its decoupled-
+ * lookback fast path — a fused custom-type scan measured 10x slower) */
+struct RankRecSize {
+    __device__ uint64_t operator()(const RankRec& r) const {
+        return (r.src & RR_KEEP) ? (uint64_t)r.full_size : 0ull;
     }
 };
-struct RankRecToAgg {
-    __device__ SurvAgg operator()(const RankRec& r) const {
-        uint64_t k = (r.src & RR_KEEP) ? 1u : 0u;
-        return {k ? (uint64_t)r.full_size : 0ull, k};
+struct RankRecFlag {
+    __device__ uint32_t operator()(const RankRec& r) const {
+        return (r.src & RR_KEEP) ? 1u : 0u;
     }
 };
 
 /* Output index records are the input format: offset u64 | key_size u32 |
  * full_size u32 (entry_writer.rs:79-87, offsets recomputed from 0).
  * src_map gets the ABSOLUTE device address of each survivor's bytes. */
-__global__ void k_emit(RunsDesc R, const RankRec* rrec, const SurvAgg* surv,
+__global__ void k_emit(RunsDesc R, const RankRec* rrec,
+                       const uint64_t* dst_off, const uint32_t* pos,
                        uint64_t total, uint8_t* out_index,
                        uint64_t* src_map) {
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
@@ -617,10 +622,9 @@ __global__ void k_emit(RunsDesc R, const RankRec* rrec, const SurvAgg* surv,
          g < total; g += stride) {
         RankRec m = rrec[g];
         if (!(m.src & RR_KEEP)) continue;
-        SurvAgg s = surv[g];
-        uint64_t p = s.cnt;
+        uint64_t p = pos[g];
         uint8_t* rec = out_index + p * 16;
-        uint64_t off = s.bytes;
+        uint64_t off = dst_off[g];
         __builtin_memcpy(rec, &off, 8);
         __builtin_memcpy(rec + 8, &m.key_size, 4);
         __builtin_memcpy(rec + 12, &m.full_size, 4);
@@ -1042,6 +1046,12 @@ done:
 /* Host side                                                          */
 /* ------------------------------------------------------------------ */
 
+static inline uint64_t ld_u64_host(const uint8_t* p) {
+    uint64_t v;
+    memcpy(&v, p, 8);
+    return v;
+}
+
 /* Max key_size across all uploaded index slabs — picks the aux tier. */
 __global__ void k_maxks(RunsDesc R, uint32_t* out) {
     __shared__ uint32_t smax;
@@ -1065,11 +1075,19 @@ __global__ void k_maxks(RunsDesc R, uint32_t* out) {
 struct dbeel_gpu_job {
     int device = -1;
     hipStream_t stream = nullptr;
+    hipStream_t copy_stream = nullptr; /* streamed-ingest H2D lane */
     hipEvent_t ev[8] = {};
+    int prep_valid = 0; /* set by job_ingest: pfx/aux/d_err already built
+                           for the current inputs (overlapped with the
+                           transfer); job_run skips the prepare stage.
+                           job_create leaves it 0 so the resident
+                           measurement mode always runs the full
+                           pipeline per step. */
     RunsDesc desc{};
     uint8_t* d_input = nullptr; /* one slab: all run data+index            */
     RankRec* d_rank = nullptr;
-    SurvAgg* d_surv = nullptr;  /* fused survivor scan output              */
+    uint64_t* d_dstoff = nullptr;
+    uint32_t* d_pos = nullptr;
     int aux_kind = 2;           /* 0: 16-B, 1: 32-B, 2: 64-B aux records   */
     uint8_t* d_outindex = nullptr;
     uint64_t* d_srcmap = nullptr;
@@ -1194,7 +1212,8 @@ extern "C" int dbeel_gpu_job_create(const dbeel_run_view* runs, size_t n_runs,
     JOB_CHECK(hipMalloc(&job->d_input, (input_bytes ? input_bytes : 16) + 32));
     uint64_t n = total ? total : 1;
     JOB_CHECK(hipMalloc(&job->d_rank, n * sizeof(RankRec)));
-    JOB_CHECK(hipMalloc(&job->d_surv, n * sizeof(SurvAgg)));
+    JOB_CHECK(hipMalloc(&job->d_dstoff, n * sizeof(uint64_t)));
+    JOB_CHECK(hipMalloc(&job->d_pos, n * sizeof(uint32_t)));
     JOB_CHECK(hipMalloc(&job->d_outindex, n * 16));
     JOB_CHECK(hipMalloc(&job->d_srcmap, n * sizeof(uint64_t)));
     JOB_CHECK(hipMalloc(&job->d_outdata, total_data ? total_data : 16));
@@ -1206,13 +1225,18 @@ extern "C" int dbeel_gpu_job_create(const dbeel_run_view* runs, size_t n_runs,
     JOB_CHECK(hipMalloc(&job->d_winp0,
                         (total_data / 8192 + 2) * sizeof(uint32_t)));
 
-    size_t t1 = 0;
+    size_t t1 = 0, t2 = 0;
     rocprim::exclusive_scan(nullptr, t1,
                             rocprim::make_transform_iterator(job->d_rank,
-                                                             RankRecToAgg{}),
-                            job->d_surv, SurvAgg{0, 0}, n, SurvAggPlus{},
-                            job->stream);
-    job->scantmp_bytes = t1;
+                                                             RankRecSize{}),
+                            job->d_dstoff, (uint64_t)0, n,
+                            rocprim::plus<uint64_t>(), job->stream);
+    rocprim::exclusive_scan(nullptr, t2,
+                            rocprim::make_transform_iterator(job->d_rank,
+                                                             RankRecFlag{}),
+                            job->d_pos, (uint32_t)0, n,
+                            rocprim::plus<uint32_t>(), job->stream);
+    job->scantmp_bytes = t1 > t2 ? t1 : t2;
     JOB_CHECK(hipMalloc(&job->d_scantmp, job->scantmp_bytes));
 
     /* Upload: one slab; record per-run device pointers. */
@@ -1316,7 +1340,8 @@ extern "C" void dbeel_gpu_job_destroy(dbeel_gpu_job* job) {
     if (job->device >= 0) hipSetDevice(job->device);
     hipFree(job->d_input);
     hipFree(job->d_rank);
-    hipFree(job->d_surv);
+    hipFree(job->d_dstoff);
+    hipFree(job->d_pos);
     hipFree(job->d_outindex);
     hipFree(job->d_srcmap);
     hipFree(job->d_outdata);
@@ -1330,7 +1355,220 @@ extern "C" void dbeel_gpu_job_destroy(dbeel_gpu_job* job) {
     for (int i = 0; i < 8; i++)
         if (job->ev[i]) hipEventDestroy(job->ev[i]);
     if (job->stream) hipStreamDestroy(job->stream);
+    if (job->copy_stream) hipStreamDestroy(job->copy_stream);
     delete job;
+}
+
+/* ------------------------------------------------------------------ */
+/* Streamed pinned ingest (north_star: pinned host DRAM, chunked       */
+/* hipMemcpyAsync, compute overlap). See include/dbeel_gpu.h.          */
+/* ------------------------------------------------------------------ */
+
+extern "C" int dbeel_gpu_pin_host(const void* ptr, size_t len) {
+    g_err[0] = 0;
+    if (!ptr || !len) {
+        set_err("pin_host: null/empty buffer");
+        return DBEEL_ERR_INVALID_ARG;
+    }
+    hipError_t e = hipHostRegister(const_cast<void*>(ptr), len,
+                                   hipHostRegisterDefault);
+    if (e == hipErrorHostMemoryAlreadyRegistered) return DBEEL_OK;
+    if (e != hipSuccess) {
+        set_err("hipHostRegister failed: %s", hipGetErrorString(e));
+        return DBEEL_ERR_HIP;
+    }
+    return DBEEL_OK;
+}
+
+extern "C" int dbeel_gpu_unpin_host(const void* ptr) {
+    g_err[0] = 0;
+    hipError_t e = hipHostUnregister(const_cast<void*>(ptr));
+    if (e != hipSuccess && e != hipErrorHostMemoryNotRegistered) {
+        set_err("hipHostUnregister failed: %s", hipGetErrorString(e));
+        return DBEEL_ERR_HIP;
+    }
+    return DBEEL_OK;
+}
+
+extern "C" int dbeel_gpu_job_ingest(dbeel_gpu_job* job,
+                                    const dbeel_run_view* runs,
+                                    size_t n_runs,
+                                    dbeel_ingest_stats* stats) {
+    g_err[0] = 0;
+    if (!job || !runs) {
+        set_err("job_ingest: null argument");
+        return DBEEL_ERR_INVALID_ARG;
+    }
+    if ((int)n_runs != job->desc.n_runs) {
+        set_err("job_ingest: run count %zu != job's %d", n_runs,
+                job->desc.n_runs);
+        return DBEEL_ERR_INVALID_ARG;
+    }
+    int rc = validate_runs(runs, n_runs);
+    if (rc) return rc;
+    for (size_t r = 0; r < n_runs; r++) {
+        if (runs[r].data_len != job->desc.data_len[r] ||
+            runs[r].index_len / 16 != job->desc.count[r]) {
+            set_err("job_ingest: run %zu shape differs from the job's", r);
+            return DBEEL_ERR_INVALID_ARG;
+        }
+    }
+    HIP_CHECK(hipSetDevice(job->device));
+    if (!job->copy_stream) HIP_CHECK(hipStreamCreate(&job->copy_stream));
+    hipStream_t cs = job->copy_stream; /* H2D lane */
+    hipStream_t ks = job->stream;      /* compute lane */
+    job->prep_valid = 0;
+    job->have_result = false;
+
+    uint64_t chunk_bytes = 64ull << 20;
+    if (const char* c = getenv("DBEEL_STREAM_CHUNK_MB")) {
+        long v = atol(c);
+        if (v >= 1) chunk_bytes = (uint64_t)v << 20;
+    }
+
+    /* chunk events: copies are in-order on cs, so event i implies all
+     * earlier chunks arrived */
+    std::vector<hipEvent_t> evs;
+    hipEvent_t ev_cs0 = nullptr, ev_cs1 = nullptr, ev_ks0 = nullptr,
+               ev_ks1 = nullptr;
+    HIP_CHECK(hipEventCreate(&ev_cs0));
+    HIP_CHECK(hipEventCreate(&ev_cs1));
+    HIP_CHECK(hipEventCreate(&ev_ks0));
+    HIP_CHECK(hipEventCreate(&ev_ks1));
+    HIP_CHECK(hipMemsetAsync(job->d_err, 0, 8, ks));
+
+    auto cleanup = [&]() {
+        for (hipEvent_t e : evs) hipEventDestroy(e);
+        hipEventDestroy(ev_cs0);
+        hipEventDestroy(ev_cs1);
+        hipEventDestroy(ev_ks0);
+        hipEventDestroy(ev_ks1);
+    };
+#define ING_CHECK(call)                                                     \
+    do {                                                                    \
+        hipError_t _e = (call);                                             \
+        if (_e != hipSuccess) {                                             \
+            set_err("%s failed: %s", #call, hipGetErrorString(_e));         \
+            cleanup();                                                      \
+            return (_e == hipErrorOutOfMemory) ? DBEEL_ERR_OOM              \
+                                               : DBEEL_ERR_HIP;             \
+        }                                                                   \
+    } while (0)
+
+    ING_CHECK(hipEventRecord(ev_cs0, cs));
+    ING_CHECK(hipEventRecord(ev_ks0, ks));
+    uint64_t total_chunks = 0;
+
+    /* index slabs first (small; prepare reads them for every entry) */
+    for (size_t r = 0; r < n_runs; r++) {
+        if (!runs[r].index_len) continue;
+        ING_CHECK(hipMemcpyAsync(
+            const_cast<uint8_t*>(job->desc.index[r]), runs[r].index,
+            runs[r].index_len, hipMemcpyHostToDevice, cs));
+    }
+    {
+        hipEvent_t e;
+        ING_CHECK(hipEventCreate(&e));
+        evs.push_back(e);
+        ING_CHECK(hipEventRecord(e, cs));
+        ING_CHECK(hipStreamWaitEvent(ks, e, 0));
+    }
+
+    /* stream each run's data in entry-aligned chunks; launch the ranged
+     * prepare for a chunk's entries as soon as its bytes are on device */
+    for (size_t r = 0; r < n_runs; r++) {
+        uint64_t n = job->desc.count[r];
+        uint64_t dlen = runs[r].data_len;
+        if (!dlen) continue;
+        const uint8_t* hidx = runs[r].index;
+        uint64_t e0 = 0, b0 = 0;
+        while (b0 < dlen) {
+            uint64_t btarget = b0 + chunk_bytes;
+            uint64_t e1, b1;
+            if (btarget >= dlen) {
+                e1 = n;
+                b1 = dlen;
+            } else {
+                /* first entry with offset >= btarget (offsets monotone,
+                 * non-overlapping — verified on-device by k_prepare) */
+                uint64_t lo = e0, hi = n;
+                while (lo < hi) {
+                    uint64_t mid = (lo + hi) >> 1;
+                    if (ld_u64_host(hidx + mid * 16) < btarget)
+                        lo = mid + 1;
+                    else
+                        hi = mid;
+                }
+                e1 = lo;
+                b1 = (e1 >= n) ? dlen : ld_u64_host(hidx + e1 * 16);
+                if (e1 >= n) e1 = n;
+                if (b1 <= b0) { /* one giant entry spans the chunk */
+                    e1 = e0 + 1;
+                    b1 = (e1 >= n) ? dlen : ld_u64_host(hidx + e1 * 16);
+                }
+            }
+            ING_CHECK(hipMemcpyAsync(
+                const_cast<uint8_t*>(job->desc.data[r]) + b0,
+                runs[r].data + b0, b1 - b0, hipMemcpyHostToDevice, cs));
+            hipEvent_t e;
+            ING_CHECK(hipEventCreate(&e));
+            evs.push_back(e);
+            ING_CHECK(hipEventRecord(e, cs));
+            ING_CHECK(hipStreamWaitEvent(ks, e, 0));
+            if (e1 > e0) {
+                uint64_t g0 = job->desc.entry_base[r] + e0;
+                uint64_t g1 = job->desc.entry_base[r] + e1;
+                uint32_t grid = pick_grid(g1 - g0, 256);
+                switch (job->aux_kind) {
+#define PREP_RANGE(KB)                                                      \
+    case (KB == 12 ? 0 : (KB == 28 ? 1 : 2)):                               \
+        hipLaunchKernelGGL(k_prepare<KB>, dim3(grid), dim3(256), 0, ks,     \
+                           job->desc, g0, g1, job->d_pfx,                   \
+                           (AuxT<KB>*)job->d_aux, job->d_err);              \
+        break;
+                    PREP_RANGE(12)
+                    PREP_RANGE(28)
+                    PREP_RANGE(60)
+#undef PREP_RANGE
+                }
+            }
+            total_chunks++;
+            e0 = e1;
+            b0 = b1;
+        }
+    }
+
+    ING_CHECK(hipEventRecord(ev_cs1, cs));
+    ING_CHECK(hipEventRecord(ev_ks1, ks));
+    uint32_t err = 0;
+    ING_CHECK(hipMemcpyAsync(&err, job->d_err, 4, hipMemcpyDeviceToHost,
+                             ks));
+    ING_CHECK(hipStreamSynchronize(cs));
+    ING_CHECK(hipStreamSynchronize(ks));
+    ING_CHECK(hipGetLastError());
+    if (err) {
+        cleanup();
+        set_err("corrupt entry/index record in streamed ingest");
+        return DBEEL_ERR_CORRUPT;
+    }
+
+    if (stats) {
+        float copy_ms = 0, prep_ms = 0, wall_ms = 0;
+        (void)hipEventElapsedTime(&copy_ms, ev_cs0, ev_cs1);
+        (void)hipEventElapsedTime(&prep_ms, ev_ks0, ev_ks1);
+        (void)hipEventElapsedTime(&wall_ms, ev_cs0, ev_ks1);
+        if (prep_ms > wall_ms) wall_ms = prep_ms;
+        stats->ingest_ms = wall_ms;
+        stats->copy_ms = copy_ms;
+        stats->prep_ms = prep_ms;
+        stats->bytes = job->input_bytes;
+        stats->chunks = total_chunks;
+    }
+    job->h2d_ms = 0.0; /* transfer accounted in ingest stats */
+    job->prep_valid = 1;
+    cleanup();
+    return DBEEL_OK;
+#undef ING_CHECK
 }
 
 static uint32_t pick_grid(uint64_t work_items, uint32_t block) {
@@ -1355,6 +1593,7 @@ extern "C" int dbeel_gpu_job_run(dbeel_gpu_job* job, int keep_tombstones,
     uint64_t n = job->total_entries;
     job->have_result = false;
 
+    int skip_prep = job->prep_valid;
     HIP_CHECK(hipMemsetAsync(job->d_err, 0, 2 * sizeof(uint32_t), s));
 
     HIP_CHECK(hipEventRecord(job->ev[0], s));
@@ -1365,9 +1604,10 @@ extern "C" int dbeel_gpu_job_run(dbeel_gpu_job* job, int keep_tombstones,
         switch (job->aux_kind) {
 #define STAGE1(KB)                                                          \
     case (KB == 12 ? 0 : (KB == 28 ? 1 : 2)):                               \
-        hipLaunchKernelGGL(k_prepare<KB>, dim3(grid), dim3(256), 0, s,      \
-                           job->desc, job->d_pfx, (AuxT<KB>*)job->d_aux,    \
-                           job->d_err);                                     \
+        if (!skip_prep)                                                     \
+            hipLaunchKernelGGL(k_prepare<KB>, dim3(grid), dim3(256), 0, s,  \
+                               job->desc, (uint64_t)0, n, job->d_pfx,       \
+                               (AuxT<KB>*)job->d_aux, job->d_err);          \
         hipEventRecord(job->ev[6], s);                                      \
         if (job->n_pairs)                                                   \
             hipLaunchKernelGGL(k_corank<KB>, dim3((uint32_t)cgrid),         \
@@ -1393,26 +1633,34 @@ extern "C" int dbeel_gpu_job_run(dbeel_gpu_job* job, int keep_tombstones,
         size_t tmp = job->scantmp_bytes;
         (void)rocprim::exclusive_scan(
             job->d_scantmp, tmp,
-            rocprim::make_transform_iterator(job->d_rank, RankRecToAgg{}),
-            job->d_surv, SurvAgg{0, 0}, n, SurvAggPlus{}, s);
+            rocprim::make_transform_iterator(job->d_rank, RankRecSize{}),
+            job->d_dstoff, (uint64_t)0, n, rocprim::plus<uint64_t>(), s);
+        tmp = job->scantmp_bytes;
+        (void)rocprim::exclusive_scan(
+            job->d_scantmp, tmp,
+            rocprim::make_transform_iterator(job->d_rank, RankRecFlag{}),
+            job->d_pos, (uint32_t)0, n, rocprim::plus<uint32_t>(), s);
     }
     HIP_CHECK(hipEventRecord(job->ev[2], s));
     if (n) {
         uint32_t grid = pick_grid(n, 256);
         hipLaunchKernelGGL(k_emit, dim3(grid), dim3(256), 0, s, job->desc,
-                           job->d_rank, job->d_surv, n, job->d_outindex,
-                           job->d_srcmap);
+                           job->d_rank, job->d_dstoff, job->d_pos, n,
+                           job->d_outindex, job->d_srcmap);
     }
     HIP_CHECK(hipEventRecord(job->ev[3], s));
 
     /* Need totals on host to size/launch the copy; one small sync. */
     RankRec last_rec = {};
-    SurvAgg last_agg = {};
+    uint64_t last_off = 0;
+    uint32_t last_pos = 0;
     uint32_t err = 0;
     if (n) {
         HIP_CHECK(hipMemcpyAsync(&last_rec, job->d_rank + (n - 1), 16,
                                  hipMemcpyDeviceToHost, s));
-        HIP_CHECK(hipMemcpyAsync(&last_agg, job->d_surv + (n - 1), 16,
+        HIP_CHECK(hipMemcpyAsync(&last_off, job->d_dstoff + (n - 1), 8,
+                                 hipMemcpyDeviceToHost, s));
+        HIP_CHECK(hipMemcpyAsync(&last_pos, job->d_pos + (n - 1), 4,
                                  hipMemcpyDeviceToHost, s));
     }
     HIP_CHECK(hipMemcpyAsync(&err, job->d_err, 4, hipMemcpyDeviceToHost, s));
@@ -1425,9 +1673,8 @@ extern "C" int dbeel_gpu_job_run(dbeel_gpu_job* job, int keep_tombstones,
         return DBEEL_ERR_CORRUPT;
     }
     uint64_t last_kept = (n && (last_rec.src & RR_KEEP)) ? 1 : 0;
-    uint64_t total_out =
-        last_agg.bytes + (last_kept ? last_rec.full_size : 0);
-    uint64_t n_surv = last_agg.cnt + last_kept;
+    uint64_t total_out = last_off + (last_kept ? last_rec.full_size : 0);
+    uint64_t n_surv = (uint64_t)last_pos + last_kept;
 
     HIP_CHECK(hipEventRecord(job->ev[4], s));
     if (total_out) {
@@ -1753,6 +2000,190 @@ extern "C" int dbeel_gpu_compact(const dbeel_run_view* runs, size_t n_runs,
                                  dbeel_compact_result* out) {
     return dbeel_gpu_compact_timed(runs, n_runs, keep_tombstones, device, out,
                                    nullptr);
+}
+
+/* ------------------------------------------------------------------ */
+/* Sliced compaction: jobs whose inputs exceed HBM                    */
+/*                                                                    */
+/* The key space is cut at pivot keys into slices whose input fits    */
+/* the device budget; each slice is an ordinary compaction job and    */
+/* the outputs concatenate. Because slice boundaries are strict key   */
+/* pivots, an equal-key group never splits, so winner/tombstone       */
+/* decisions are slice-local and the concatenation is byte-identical  */
+/* to one whole compaction (survivors verbatim in global key order,   */
+/* offsets rebased). Requires densely-ordered runs (EntryWriter       */
+/* layout — verified on-device).                                      */
+/* ------------------------------------------------------------------ */
+
+static int host_key_cmp(const uint8_t* a, uint64_t la, const uint8_t* b,
+                        uint64_t lb) {
+    uint64_t n = la < lb ? la : lb;
+    int c = memcmp(a, b, n);
+    if (c) return c;
+    return la < lb ? -1 : (la > lb ? 1 : 0);
+}
+
+/* key view of entry i of a host run */
+static inline const uint8_t* host_entry_key(const dbeel_run_view* run,
+                                            uint64_t i, uint64_t* klen) {
+    const uint8_t* rec = run->index + i * 16;
+    uint64_t off = ld_u64_host(rec);
+    uint32_t key_size;
+    memcpy(&key_size, rec + 8, 4);
+    *klen = key_size - 8;
+    return run->data + off + 8;
+}
+
+extern "C" int dbeel_gpu_compact_sliced(const dbeel_run_view* runs,
+                                        size_t n_runs, int keep_tombstones,
+                                        int device,
+                                        uint64_t max_resident_bytes,
+                                        dbeel_compact_result* out) {
+    g_err[0] = 0;
+    if (!out) {
+        set_err("out is null");
+        return DBEEL_ERR_INVALID_ARG;
+    }
+    int rc = validate_runs(runs, n_runs);
+    if (rc) return rc;
+    if (device < 0) {
+        set_err("device must be >= 0 (no CPU fallback)");
+        return DBEEL_ERR_INVALID_ARG;
+    }
+    uint64_t total_input = 0, max_count = 0;
+    size_t big_run = 0;
+    for (size_t r = 0; r < n_runs; r++) {
+        total_input += runs[r].data_len + runs[r].index_len;
+        uint64_t c = runs[r].index_len / 16;
+        if (c > max_count) {
+            max_count = c;
+            big_run = r;
+        }
+    }
+    uint64_t budget = max_resident_bytes;
+    if (!budget) {
+        HIP_CHECK(hipSetDevice(device));
+        size_t free_b = 0, total_b = 0;
+        HIP_CHECK(hipMemGetInfo(&free_b, &total_b));
+        /* input slab + output data + pfx/aux/cr/rrec intermediates stay
+         * under ~2.6x input for <= 64 runs; /3 leaves headroom */
+        budget = (uint64_t)(free_b / 3);
+    }
+    uint64_t S = budget ? (total_input + budget - 1) / budget : 1;
+    if (S <= 1 || max_count < 2 * S)
+        return dbeel_gpu_compact(runs, n_runs, keep_tombstones, device, out);
+
+    /* pivots: evenly spaced keys of the largest run */
+    std::vector<std::pair<const uint8_t*, uint64_t>> pivots; /* key,len */
+    for (uint64_t j = 1; j < S; j++) {
+        uint64_t klen;
+        const uint8_t* k =
+            host_entry_key(&runs[big_run], j * max_count / S, &klen);
+        pivots.push_back({k, klen});
+    }
+    /* per run: slice boundary entry indices (first entry >= pivot) */
+    std::vector<std::vector<uint64_t>> bounds(n_runs);
+    for (size_t r = 0; r < n_runs; r++) {
+        uint64_t n = runs[r].index_len / 16;
+        bounds[r].push_back(0);
+        for (auto& pv : pivots) {
+            uint64_t lo = bounds[r].back(), hi = n;
+            while (lo < hi) {
+                uint64_t mid = (lo + hi) >> 1;
+                uint64_t kl;
+                const uint8_t* k = host_entry_key(&runs[r], mid, &kl);
+                if (host_key_cmp(k, kl, pv.first, pv.second) < 0)
+                    lo = mid + 1;
+                else
+                    hi = mid;
+            }
+            bounds[r].push_back(lo);
+        }
+        bounds[r].push_back(n);
+    }
+
+    memset(out, 0, sizeof *out);
+    std::vector<uint8_t> acc_data, acc_index;
+    uint64_t written = 0;
+    std::vector<uint8_t> idx_tmp;
+    for (uint64_t sl = 0; sl < S; sl++) {
+        std::vector<dbeel_run_view> views;
+        std::vector<size_t> idx_tmp_off;
+        idx_tmp.clear();
+        uint64_t slice_entries = 0;
+        for (size_t r = 0; r < n_runs; r++) {
+            uint64_t lo = bounds[r][sl], hi = bounds[r][sl + 1];
+            dbeel_run_view v{};
+            if (hi > lo) {
+                uint64_t off_lo = ld_u64_host(runs[r].index + lo * 16);
+                uint64_t off_hi =
+                    (hi < runs[r].index_len / 16)
+                        ? ld_u64_host(runs[r].index + hi * 16)
+                        : runs[r].data_len;
+                v.data = runs[r].data + off_lo;
+                v.data_len = off_hi - off_lo;
+                /* rebased index records */
+                size_t pos = idx_tmp.size();
+                idx_tmp.resize(pos + (hi - lo) * 16);
+                memcpy(idx_tmp.data() + pos, runs[r].index + lo * 16,
+                       (hi - lo) * 16);
+                for (uint64_t i = 0; i < hi - lo; i++) {
+                    uint64_t o =
+                        ld_u64_host(idx_tmp.data() + pos + i * 16) - off_lo;
+                    memcpy(idx_tmp.data() + pos + i * 16, &o, 8);
+                }
+                idx_tmp_off.push_back(pos);
+                v.index_len = (hi - lo) * 16;
+                slice_entries += hi - lo;
+            } else {
+                v.data = runs[r].data; /* non-null, zero-length */
+                v.data_len = 0;
+                v.index = runs[r].index;
+                v.index_len = 0;
+                idx_tmp_off.push_back((size_t)-1);
+            }
+            views.push_back(v);
+        }
+        if (!slice_entries) continue;
+        /* resolve idx pointers after idx_tmp stopped growing */
+        for (size_t r = 0; r < n_runs; r++)
+            if (idx_tmp_off[r] != (size_t)-1)
+                views[r].index = idx_tmp.data() + idx_tmp_off[r];
+        dbeel_compact_result sres{};
+        rc = dbeel_gpu_compact(views.data(), n_runs, keep_tombstones,
+                               device, &sres);
+        if (rc) return rc;
+        /* concatenate, rebasing index offsets */
+        uint64_t base = acc_data.size();
+        acc_data.insert(acc_data.end(), sres.data,
+                        sres.data + sres.data_len);
+        size_t ipos = acc_index.size();
+        acc_index.insert(acc_index.end(), sres.index,
+                         sres.index + sres.index_len);
+        for (uint64_t i = 0; i < sres.index_len / 16; i++) {
+            uint64_t o =
+                ld_u64_host(acc_index.data() + ipos + i * 16) + base;
+            memcpy(acc_index.data() + ipos + i * 16, &o, 8);
+        }
+        written += sres.entries_written;
+        dbeel_gpu_result_free(&sres);
+    }
+
+    out->data = (uint8_t*)malloc(acc_data.size() ? acc_data.size() : 1);
+    out->index = (uint8_t*)malloc(acc_index.size() ? acc_index.size() : 1);
+    if (!out->data || !out->index) {
+        free(out->data);
+        free(out->index);
+        memset(out, 0, sizeof *out);
+        set_err("host alloc failed");
+        return DBEEL_ERR_OOM;
+    }
+    memcpy(out->data, acc_data.data(), acc_data.size());
+    memcpy(out->index, acc_index.data(), acc_index.size());
+    out->data_len = acc_data.size();
+    out->index_len = acc_index.size();
+    out->entries_written = written;
+    return DBEEL_OK;
 }
 
 extern "C" void dbeel_gpu_result_free(dbeel_compact_result* r) {

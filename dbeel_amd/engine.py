@@ -121,6 +121,20 @@ def load() -> ctypes.CDLL:
             ctypes.POINTER(ctypes.c_uint8), ctypes.c_int,
             ctypes.POINTER(CompactResult),
         ]
+        lib.dbeel_gpu_pin_host.restype = ctypes.c_int
+        lib.dbeel_gpu_pin_host.argtypes = [ctypes.c_void_p, ctypes.c_size_t]
+        lib.dbeel_gpu_unpin_host.restype = ctypes.c_int
+        lib.dbeel_gpu_unpin_host.argtypes = [ctypes.c_void_p]
+        lib.dbeel_gpu_job_ingest.restype = ctypes.c_int
+        lib.dbeel_gpu_job_ingest.argtypes = [
+            ctypes.c_void_p, ctypes.POINTER(RunView), ctypes.c_size_t,
+            ctypes.POINTER(IngestStats),
+        ]
+        lib.dbeel_gpu_compact_sliced.restype = ctypes.c_int
+        lib.dbeel_gpu_compact_sliced.argtypes = [
+            ctypes.POINTER(RunView), ctypes.c_size_t, ctypes.c_int,
+            ctypes.c_int, ctypes.c_uint64, ctypes.POINTER(CompactResult),
+        ]
         _lib = lib
     return _lib
 
@@ -170,6 +184,59 @@ def compact(runs, keep_tombstones: bool, device: int = 0,
     del keepalive
     if want_timings:
         return data, index, n, tim.as_dict()
+    return data, index, n
+
+
+class IngestStats(ctypes.Structure):
+    _fields_ = [
+        ("ingest_ms", ctypes.c_double),
+        ("copy_ms", ctypes.c_double),
+        ("prep_ms", ctypes.c_double),
+        ("bytes", ctypes.c_uint64),
+        ("chunks", ctypes.c_uint64),
+    ]
+
+    def as_dict(self):
+        return {k: getattr(self, k) for k, _ in self._fields_}
+
+
+def pin_host(arr: np.ndarray) -> None:
+    """Page-lock a numpy buffer (hipHostRegister) so streamed ingest runs
+    as true async DMA. Pin once, outside the hot loop."""
+    lib = load()
+    rc = lib.dbeel_gpu_pin_host(arr.ctypes.data, arr.nbytes)
+    if rc != 0:
+        raise DbeelGpuError(rc, lib.dbeel_gpu_last_error().decode())
+
+
+def unpin_host(arr: np.ndarray) -> None:
+    lib = load()
+    rc = lib.dbeel_gpu_unpin_host(arr.ctypes.data)
+    if rc != 0:
+        raise DbeelGpuError(rc, lib.dbeel_gpu_last_error().decode())
+
+
+def compact_sliced(runs, keep_tombstones: bool, device: int = 0,
+                   max_resident_bytes: int = 0):
+    """Sliced compaction for inputs larger than HBM: key space cut at
+    pivots, each slice compacted on-device, outputs concatenated —
+    byte-identical to one whole compaction (include/dbeel_gpu.h)."""
+    lib = load()
+    views, keepalive = _views(runs)
+    res = CompactResult()
+    rc = lib.dbeel_gpu_compact_sliced(
+        views, len(runs), int(keep_tombstones), device,
+        max_resident_bytes, ctypes.byref(res),
+    )
+    if rc != 0:
+        raise DbeelGpuError(rc, lib.dbeel_gpu_last_error().decode())
+    try:
+        data = _ptr_bytes(res.data, res.data_len)
+        index = _ptr_bytes(res.index, res.index_len)
+        n = int(res.entries_written)
+    finally:
+        lib.dbeel_gpu_result_free(ctypes.byref(res))
+    del keepalive
     return data, index, n
 
 
@@ -294,6 +361,23 @@ class Job:
         if rc != 0:
             raise DbeelGpuError(rc, self._lib.dbeel_gpu_last_error().decode())
         return int(dl.value), int(ne.value), tim.as_dict()
+
+    def ingest(self, runs):
+        """Streamed pinned ingest (north_star): re-upload fresh run
+        contents into this job's resident slab, chunked hipMemcpyAsync on
+        a copy stream overlapped with the prepare kernel on the compute
+        stream. Shapes must match the job's. The next run() skips the
+        prepare stage (already done, hidden behind the transfer).
+        Returns the ingest stats dict."""
+        views, keepalive = _views(runs)
+        st = IngestStats()
+        rc = self._lib.dbeel_gpu_job_ingest(
+            self._h, views, len(views), ctypes.byref(st)
+        )
+        if rc != 0:
+            raise DbeelGpuError(rc, self._lib.dbeel_gpu_last_error().decode())
+        del keepalive
+        return st.as_dict()
 
     def fetch(self):
         res = CompactResult()

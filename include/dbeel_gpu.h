@@ -175,6 +175,52 @@ int dbeel_gpu_job_bloom(dbeel_gpu_job* job, uint8_t** out_bytes,
 void dbeel_gpu_bloom_free(uint8_t* p);
 void dbeel_gpu_job_destroy(dbeel_gpu_job* job);
 
+/* ---- Streamed pinned ingest (north_star: "input SSTable runs are
+ * pinned in host DRAM and streamed via hipMemcpyAsync with compute
+ * overlap") ----
+ *
+ * dbeel_gpu_pin_host / unpin_host: page-lock a caller buffer
+ * (hipHostRegister) so streamed copies run as true async DMA. A
+ * production engine pins its run staging buffers once; pin before the
+ * hot loop, not inside it.
+ *
+ * dbeel_gpu_job_ingest: re-uploads fresh run contents into an EXISTING
+ * job's resident input slab, streaming each run's data in
+ * entry-boundary-aligned chunks (default 64 MiB, env
+ * DBEEL_STREAM_CHUNK_MB) on a dedicated copy stream while k_prepare
+ * runs on already-arrived chunks on the compute stream — the
+ * copy-engine/compute overlap the reference gets from its 16-deep
+ * read-ahead DmaStreamReaders (lsm_tree.rs:974-993). Run shapes
+ * (data_len/index_len per run) must match the job's. After a successful
+ * ingest the job's prepare stage is already done: the next
+ * dbeel_gpu_job_run skips it (the work happened, hidden behind the
+ * PCIe transfer). stats (may be NULL) reports the wall ingest time and
+ * how much prepare time was hidden. */
+typedef struct {
+    double ingest_ms;   /* wall: first copy enqueued -> all chunks +
+                           prepare complete */
+    double copy_ms;     /* pure H2D transfer time (copy stream span) */
+    double prep_ms;     /* prepare kernel time, overlapped with copies */
+    uint64_t bytes;     /* total bytes ingested (data + index) */
+    uint64_t chunks;
+} dbeel_ingest_stats;
+
+int dbeel_gpu_pin_host(const void* ptr, size_t len);
+int dbeel_gpu_unpin_host(const void* ptr);
+int dbeel_gpu_job_ingest(dbeel_gpu_job* job, const dbeel_run_view* runs,
+                         size_t n_runs, dbeel_ingest_stats* stats);
+
+/* ---- Sliced compaction (runs larger than HBM) ----
+ * Splits the key space into slices that fit max_resident_bytes of input,
+ * compacts each slice independently (equal keys never split: slice
+ * boundaries are strict key pivots) and concatenates the outputs —
+ * byte-identical to one whole-job compaction of the same runs.
+ * max_resident_bytes = 0 picks a default from free device memory. */
+int dbeel_gpu_compact_sliced(const dbeel_run_view* runs, size_t n_runs,
+                             int keep_tombstones, int device,
+                             uint64_t max_resident_bytes,
+                             dbeel_compact_result* out);
+
 #ifdef __cplusplus
 }
 #endif

@@ -13,13 +13,17 @@ LIB = os.path.join(REPO_ROOT, "dbeel_amd", "libdbeel_gpu.so")
 HEADER_SYMBOLS = [
     "dbeel_gpu_compact",
     "dbeel_gpu_compact_timed",
+    "dbeel_gpu_compact_sliced",
     "dbeel_gpu_result_free",
     "dbeel_gpu_last_error",
     "dbeel_gpu_encode_run",
     "dbeel_gpu_job_create",
     "dbeel_gpu_job_run",
     "dbeel_gpu_job_fetch",
+    "dbeel_gpu_job_ingest",
     "dbeel_gpu_job_destroy",
+    "dbeel_gpu_pin_host",
+    "dbeel_gpu_unpin_host",
 ]
 
 LSM_SYMBOLS = [
