@@ -1978,6 +1978,240 @@ extern "C" int dbeel_gpu_lookup(const dbeel_run_view* runs, size_t n_runs,
     return DBEEL_OK;
 }
 
+/* ------------------------------------------------------------------ */
+/* Migration / iteration scan (AsyncIter analogue)                    */
+/*                                                                    */
+/* See include/dbeel_gpu.h. The yield order is the reference's:       */
+/* sstables ascending, entries in index order within each             */
+/* (lsm_tree.rs:210-276) — which is exactly the global entry order g, */
+/* so the existing survivor-scan/emit/copy machinery packs the        */
+/* filtered entries verbatim. No dedup, no tombstone filter.          */
+/* ------------------------------------------------------------------ */
+
+/* MurmurHash3 x86_32 — restatement of the murmur3 crate v0.5.2's      */
+/* murmur3_32 (the reference's hash_bytes, shards.rs:96-101, seed 0   */
+/* everywhere). Public algorithm (Appleby); pinned by the published   */
+/* test vectors in tests/test_scan_host.py.                           */
+__device__ __host__ __forceinline__ uint32_t rotl32(uint32_t x, int r) {
+    return (x << r) | (x >> (32 - r));
+}
+__device__ __host__ inline uint32_t murmur3_32(const uint8_t* key,
+                                               uint64_t len,
+                                               uint32_t seed) {
+    const uint32_t c1 = 0xcc9e2d51u, c2 = 0x1b873593u;
+    uint32_t h = seed;
+    uint64_t nblocks = len / 4;
+    for (uint64_t i = 0; i < nblocks; i++) {
+        uint32_t k;
+        __builtin_memcpy(&k, key + i * 4, 4); /* little-endian */
+        k *= c1;
+        k = rotl32(k, 15);
+        k *= c2;
+        h ^= k;
+        h = rotl32(h, 13);
+        h = h * 5 + 0xe6546b64u;
+    }
+    uint32_t k1 = 0;
+    const uint8_t* tail = key + nblocks * 4;
+    switch (len & 3) {
+        case 3: k1 ^= (uint32_t)tail[2] << 16; /* fallthrough */
+        case 2: k1 ^= (uint32_t)tail[1] << 8;  /* fallthrough */
+        case 1:
+            k1 ^= tail[0];
+            k1 *= c1;
+            k1 = rotl32(k1, 15);
+            k1 *= c2;
+            h ^= k1;
+    }
+    h ^= (uint32_t)len;
+    h ^= h >> 16;
+    h *= 0x85ebca6bu;
+    h ^= h >> 13;
+    h *= 0xc2b2ae35u;
+    h ^= h >> 16;
+    return h;
+}
+
+/* host-callable export so the CPU test suite can pin the restatement
+ * against the published MurmurHash3 vectors without a GPU */
+extern "C" uint32_t dbeel_murmur3_32(const uint8_t* key, uint64_t len,
+                                     uint32_t seed) {
+    return murmur3_32(key, len, seed);
+}
+
+/* between_cmp restated EXACTLY (tasks/migration.rs:54-60). Note the
+ * reference's wrapped case (end < start) evaluates true for every hash —
+ * restated verbatim, divergence-free. */
+__device__ __forceinline__ bool hash_between(uint32_t h, uint32_t start,
+                                             uint32_t end) {
+    if (end < start) return h < start || h >= end;
+    return h >= start && h < end;
+}
+
+__global__ void k_scanflag(RunsDesc R, const uint8_t* kbounds,
+                           uint64_t start_len, uint64_t end_len,
+                           int have_start, int have_end,
+                           const uint32_t* rstart, const uint32_t* rend,
+                           uint32_t n_ranges, RankRec* rrec,
+                           uint32_t* err) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t g = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         g < R.total; g += stride) {
+        int r = 0;
+        while (r + 1 < R.n_runs && g >= R.entry_base[r + 1]) r++;
+        uint64_t i = g - R.entry_base[r];
+        EView e;
+        if (!load_entry(R, r, i, e)) {
+            atomicOr(err, DERR_CORRUPT);
+            RankRec z = {0, 8, 32};
+            rrec[g] = z;
+            continue;
+        }
+        bool keep = true;
+        if (have_start &&
+            cmp_keys(e.key, e.klen, kbounds, start_len) < 0)
+            keep = false;
+        if (keep && have_end &&
+            cmp_keys(e.key, e.klen, kbounds + start_len, end_len) >= 0)
+            keep = false;
+        if (keep && n_ranges) {
+            uint32_t h = murmur3_32(e.key, e.klen, 0);
+            bool in_any = false;
+            for (uint32_t j = 0; j < n_ranges && !in_any; j++)
+                in_any = hash_between(h, rstart[j], rend[j]);
+            keep = in_any;
+        }
+        RankRec m;
+        m.src = ((uint64_t)r << 48) | e.off | (keep ? RR_KEEP : 0);
+        m.key_size = e.key_size;
+        m.full_size = e.full_size;
+        rrec[g] = m; /* dense: rank == g (yield order, no reordering) */
+    }
+}
+
+extern "C" int dbeel_gpu_scan(const dbeel_run_view* runs, size_t n_runs,
+                              const uint8_t* start_key,
+                              size_t start_key_len, const uint8_t* end_key,
+                              size_t end_key_len,
+                              const uint32_t* range_starts,
+                              const uint32_t* range_ends, size_t n_ranges,
+                              int device, dbeel_compact_result* out) {
+    g_err[0] = 0;
+    if (!out || (n_ranges && (!range_starts || !range_ends)) ||
+        (start_key_len && !start_key) || (end_key_len && !end_key)) {
+        set_err("scan: null argument");
+        return DBEEL_ERR_INVALID_ARG;
+    }
+    dbeel_gpu_job* job = nullptr;
+    int rc = dbeel_gpu_job_create(runs, n_runs, device, &job);
+    if (rc) return rc;
+    hipStream_t s = job->stream;
+    uint64_t n = job->total_entries;
+
+    uint8_t* d_kb = nullptr;
+    uint32_t* d_ranges = nullptr;
+    hipError_t e = hipSuccess;
+    uint64_t kb_len = start_key_len + end_key_len;
+    if (kb_len == 0) kb_len = 1;
+    e = hipMalloc(&d_kb, kb_len);
+    if (e == hipSuccess && start_key_len)
+        e = hipMemcpyAsync(d_kb, start_key, start_key_len,
+                           hipMemcpyHostToDevice, s);
+    if (e == hipSuccess && end_key_len)
+        e = hipMemcpyAsync(d_kb + start_key_len, end_key, end_key_len,
+                           hipMemcpyHostToDevice, s);
+    if (e == hipSuccess)
+        e = hipMalloc(&d_ranges, (n_ranges ? 2 * n_ranges : 1) * 4);
+    if (e == hipSuccess && n_ranges) {
+        e = hipMemcpyAsync(d_ranges, range_starts, n_ranges * 4,
+                           hipMemcpyHostToDevice, s);
+        if (e == hipSuccess)
+            e = hipMemcpyAsync(d_ranges + n_ranges, range_ends,
+                               n_ranges * 4, hipMemcpyHostToDevice, s);
+    }
+    if (e == hipSuccess)
+        e = hipMemsetAsync(job->d_err, 0, 8, s);
+    if (e == hipSuccess && n) {
+        hipLaunchKernelGGL(k_scanflag, dim3(pick_grid(n, 256)), dim3(256),
+                           0, s, job->desc, d_kb, (uint64_t)start_key_len,
+                           (uint64_t)end_key_len, start_key ? 1 : 0,
+                           end_key ? 1 : 0, d_ranges,
+                           d_ranges + n_ranges, (uint32_t)n_ranges,
+                           job->d_rank, job->d_err);
+        size_t tmp = job->scantmp_bytes;
+        (void)rocprim::exclusive_scan(
+            job->d_scantmp, tmp,
+            rocprim::make_transform_iterator(job->d_rank, RankRecSize{}),
+            job->d_dstoff, (uint64_t)0, n, rocprim::plus<uint64_t>(), s);
+        tmp = job->scantmp_bytes;
+        (void)rocprim::exclusive_scan(
+            job->d_scantmp, tmp,
+            rocprim::make_transform_iterator(job->d_rank, RankRecFlag{}),
+            job->d_pos, (uint32_t)0, n, rocprim::plus<uint32_t>(), s);
+        hipLaunchKernelGGL(k_emit, dim3(pick_grid(n, 256)), dim3(256), 0,
+                           s, job->desc, job->d_rank, job->d_dstoff,
+                           job->d_pos, n, job->d_outindex, job->d_srcmap);
+    }
+    RankRec last_rec = {};
+    uint64_t last_off = 0;
+    uint32_t last_pos = 0, derr = 0;
+    if (e == hipSuccess && n) {
+        e = hipMemcpyAsync(&last_rec, job->d_rank + (n - 1), 16,
+                           hipMemcpyDeviceToHost, s);
+        if (e == hipSuccess)
+            e = hipMemcpyAsync(&last_off, job->d_dstoff + (n - 1), 8,
+                               hipMemcpyDeviceToHost, s);
+        if (e == hipSuccess)
+            e = hipMemcpyAsync(&last_pos, job->d_pos + (n - 1), 4,
+                               hipMemcpyDeviceToHost, s);
+    }
+    if (e == hipSuccess)
+        e = hipMemcpyAsync(&derr, job->d_err, 4, hipMemcpyDeviceToHost, s);
+    if (e == hipSuccess) e = hipStreamSynchronize(s);
+    if (e == hipSuccess) e = hipGetLastError();
+
+    int ret = DBEEL_OK;
+    if (e != hipSuccess) {
+        set_err("scan failed: %s", hipGetErrorString(e));
+        ret = DBEEL_ERR_HIP;
+    } else if (derr) {
+        set_err("corrupt entry/index record");
+        ret = DBEEL_ERR_CORRUPT;
+    } else {
+        uint64_t last_kept = (n && (last_rec.src & RR_KEEP)) ? 1 : 0;
+        uint64_t total_out =
+            last_off + (last_kept ? last_rec.full_size : 0);
+        uint64_t n_surv = (uint64_t)last_pos + last_kept;
+        if (total_out) {
+            uint32_t win = 16384;
+            uint64_t windows = (total_out + win - 1) / win;
+            uint32_t grid = windows > 4096 ? 4096 : (uint32_t)windows;
+            hipLaunchKernelGGL(k_winmap, dim3(pick_grid(windows, 256)),
+                               dim3(256), 0, s, job->d_outindex, n_surv,
+                               total_out, win, job->d_winp0);
+            hipLaunchKernelGGL((k_copy<256, 16384>), dim3(grid), dim3(256),
+                               0, s, job->d_outindex, job->d_srcmap,
+                               job->d_winp0, n_surv, total_out,
+                               job->d_outdata);
+        }
+        e = hipStreamSynchronize(s);
+        if (e == hipSuccess) e = hipGetLastError();
+        if (e != hipSuccess) {
+            set_err("scan copy failed: %s", hipGetErrorString(e));
+            ret = DBEEL_ERR_HIP;
+        } else {
+            job->out_data_len = total_out;
+            job->out_entries = n_surv;
+            job->have_result = true;
+            ret = dbeel_gpu_job_fetch(job, out);
+        }
+    }
+    hipFree(d_kb);
+    hipFree(d_ranges);
+    dbeel_gpu_job_destroy(job);
+    return ret;
+}
+
 extern "C" int dbeel_gpu_compact_timed(const dbeel_run_view* runs,
                                        size_t n_runs, int keep_tombstones,
                                        int device, dbeel_compact_result* out,

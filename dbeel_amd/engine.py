@@ -135,6 +135,14 @@ def load() -> ctypes.CDLL:
             ctypes.POINTER(RunView), ctypes.c_size_t, ctypes.c_int,
             ctypes.c_int, ctypes.c_uint64, ctypes.POINTER(CompactResult),
         ]
+        lib.dbeel_gpu_scan.restype = ctypes.c_int
+        lib.dbeel_gpu_scan.argtypes = [
+            ctypes.POINTER(RunView), ctypes.c_size_t,
+            ctypes.POINTER(ctypes.c_uint8), ctypes.c_size_t,
+            ctypes.POINTER(ctypes.c_uint8), ctypes.c_size_t,
+            ctypes.POINTER(ctypes.c_uint32), ctypes.POINTER(ctypes.c_uint32),
+            ctypes.c_size_t, ctypes.c_int, ctypes.POINTER(CompactResult),
+        ]
         _lib = lib
     return _lib
 
@@ -184,6 +192,49 @@ def compact(runs, keep_tombstones: bool, device: int = 0,
     del keepalive
     if want_timings:
         return data, index, n, tim.as_dict()
+    return data, index, n
+
+
+def scan(runs, start_key: bytes | None = None,
+         end_key: bytes | None = None, hash_ranges=None, device: int = 0):
+    """Migration/iteration scan (AsyncIter analogue, lsm_tree.rs:141-282 +
+    tasks/migration.rs:62-131): yields every entry in the reference's
+    order — runs ascending, entries in key order within each — with no
+    dedup and no tombstone filter, restricted by an optional key range
+    [start_key, end_key) and/or murmur3_32 hash ranges [(start, end), ...]
+    (hash_bytes shards.rs:99-101, between_cmp migration.rs:54-60).
+    Returns (data_bytes, index_bytes, n) shaped like a run file."""
+    lib = load()
+    views, keepalive = _views(runs)
+
+    def _kb(b):
+        if b is None:
+            return None, 0, None
+        a = np.frombuffer(b or b"\0", dtype=np.uint8)
+        return a.ctypes.data_as(ctypes.POINTER(ctypes.c_uint8)), len(b), a
+
+    sp, sl, sa = _kb(start_key)
+    ep, el, ea = _kb(end_key)
+    n_ranges = len(hash_ranges) if hash_ranges else 0
+    if n_ranges:
+        rs = np.array([r[0] for r in hash_ranges], dtype=np.uint32)
+        re_ = np.array([r[1] for r in hash_ranges], dtype=np.uint32)
+        rsp = rs.ctypes.data_as(ctypes.POINTER(ctypes.c_uint32))
+        rep = re_.ctypes.data_as(ctypes.POINTER(ctypes.c_uint32))
+    else:
+        rsp = rep = None
+    res = CompactResult()
+    rc = lib.dbeel_gpu_scan(views, len(runs), sp, sl, ep, el, rsp, rep,
+                            n_ranges, device, ctypes.byref(res))
+    if rc != 0:
+        raise DbeelGpuError(rc, lib.dbeel_gpu_last_error().decode())
+    try:
+        data = _ptr_bytes(res.data, res.data_len)
+        index = _ptr_bytes(res.index, res.index_len)
+        n = int(res.entries_written)
+    finally:
+        lib.dbeel_gpu_result_free(ctypes.byref(res))
+    del keepalive
     return data, index, n
 
 

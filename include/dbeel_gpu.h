@@ -210,6 +210,28 @@ int dbeel_gpu_unpin_host(const void* ptr);
 int dbeel_gpu_job_ingest(dbeel_gpu_job* job, const dbeel_run_view* runs,
                          size_t n_runs, dbeel_ingest_stats* stats);
 
+/* ---- Batched migration / iteration scan ----
+ * GPU analogue of AsyncIter (lsm_tree.rs:141-282) feeding migration
+ * (tasks/migration.rs:62-131): yields entries in the reference's yield
+ * order — sstables ascending by index, entries in key order within each
+ * (IterState walks index 0 upward, lsm_tree.rs:210-276) — with NO dedup
+ * and NO tombstone filtering (the reference iterator yields every entry;
+ * migration re-sends duplicates and resolves by timestamp at the
+ * destination). Filters, combinable (entry kept iff it passes all):
+ *   - key range [start_key, end_key), either bound NULL = unbounded;
+ *   - murmur3_32(key, seed 0) hash ranges with wraparound semantics
+ *     (hash_bytes shards.rs:99-101, murmur3 crate v0.5.2 restated;
+ *     between_cmp tasks/migration.rs:54-60: start <= end means
+ *     [start, end), end < start wraps to hash >= start || hash < end...
+ *     precisely: kept iff hash < start || hash >= end when end < start).
+ * Output = packed entries verbatim + rebuilt 16-B index records, like a
+ * run file holding exactly the yielded entries. */
+int dbeel_gpu_scan(const dbeel_run_view* runs, size_t n_runs,
+                   const uint8_t* start_key, size_t start_key_len,
+                   const uint8_t* end_key, size_t end_key_len,
+                   const uint32_t* range_starts, const uint32_t* range_ends,
+                   size_t n_ranges, int device, dbeel_compact_result* out);
+
 /* ---- Sliced compaction (runs larger than HBM) ----
  * Splits the key space into slices that fit max_resident_bytes of input,
  * compacts each slice independently (equal keys never split: slice
