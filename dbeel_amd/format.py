@@ -116,34 +116,57 @@ def build_run_fixed_key(
 
     full = (ENTRY_OVERHEAD + K + value_sizes).astype(np.uint64)
     maxw = ENTRY_OVERHEAD + K + V
-    M = np.zeros((N, maxw), dtype=np.uint8)
+    uniform = bool(np.all(value_sizes == V))
+    if uniform:
+        # no tombstones: every entry is maxw wide, so the (N, maxw)
+        # matrix IS the data file — skip the boolean-mask compact (the
+        # dominant cost at GiB scale)
+        flat = np.zeros(N * maxw, dtype=np.uint8)
+        M = flat.reshape(N, maxw)
+    else:
+        M = np.zeros((N, maxw), dtype=np.uint8)
 
-    # key_len:u64 LE
-    M[:, 0] = K & 0xFF
-    for j in range(1, 8):
-        M[:, j] = (K >> (8 * j)) & 0xFF
+    # key_len:u64 LE (constant row prefix)
+    M[:, 0:8] = np.frombuffer(
+        int(K).to_bytes(8, "little"), dtype=np.uint8
+    )[None, :]
     # key bytes
     M[:, 8 : 8 + K] = keys
-    # data_len:u64 LE at col 8+K
-    dl = value_sizes
-    for j in range(8):
-        M[:, 8 + K + j] = ((dl >> np.uint64(8 * j)) & np.uint64(0xFF)).astype(np.uint8)
-    # values + timestamp: two classes by tombstone-ness
+    # data_len:u64 LE at col 8+K (V everywhere, 0 on tombstone rows)
     is_tomb = value_sizes == 0
-    norm = ~is_tomb
+    it = np.flatnonzero(is_tomb)
+    M[:, 8 + K : 16 + K] = value_sizes.astype("<u8").view(
+        np.uint8).reshape(N, 8)
+    ts_bytes = timestamps_lo.astype("<u8").view(np.uint8).reshape(N, 8)
     if V:
-        M[norm, 16 + K : 16 + K + V] = value_fill.reshape(-1, V)
-    ts = timestamps_lo.astype(np.uint64)
-    # timestamp i128 LE: low u64 then high u64 (= 0 here)
-    for j in range(8):
-        tb = ((ts >> np.uint64(8 * j)) & np.uint64(0xFF)).astype(np.uint8)
-        if V:
-            M[norm, 16 + K + V + j] = tb[norm]
-        M[is_tomb, 16 + K + j] = tb[is_tomb]
-    # (high 8 bytes already zero)
+        # values for non-tombstone rows; timestamp written unconditionally
+        # at the full-width position (tombstone rows' copy lands past
+        # their entry end and is masked out of the compact below)
+        inorm = np.flatnonzero(~is_tomb)
+        M[inorm, 16 + K : 16 + K + V] = value_fill.reshape(-1, V)
+        M[:, 16 + K + V : 24 + K + V] = ts_bytes
+    if it.size or not V:
+        rows = it if V else np.arange(N)
+        M[rows, 16 + K : 24 + K] = ts_bytes[rows]
+    # (timestamp high 8 bytes already zero)
 
-    mask = np.arange(maxw, dtype=np.uint64)[None, :] < full[:, None]
-    data = M[mask]
+    if uniform:
+        data = flat
+    else:
+        # compact out the unused tail of tombstone rows by concatenating
+        # contiguous full-width segments + truncated tombstone rows —
+        # one copy pass, no (N, maxw) boolean mask (which alone costs
+        # ~maxw bytes/entry of extra memory and two passes)
+        pieces = []
+        prev = 0
+        for t in it:
+            if t > prev:
+                pieces.append(M[prev:t].reshape(-1))
+            pieces.append(M[t, : ENTRY_OVERHEAD + K])
+            prev = t + 1
+        if prev < N:
+            pieces.append(M[prev:].reshape(-1))
+        data = np.concatenate(pieces) if pieces else M[:0].reshape(-1)
 
     offsets = np.zeros(N, dtype=np.uint64)
     np.cumsum(full[:-1], out=offsets[1:])
