@@ -1680,12 +1680,18 @@ extern "C" int dbeel_gpu_job_run(dbeel_gpu_job* job, int keep_tombstones,
     if (total_out) {
         int variant = 0;
         if (const char* v = getenv("DBEEL_COPY_VARIANT")) variant = atoi(v);
-        uint32_t win = (variant == 1) ? 8192
+        uint32_t win = (variant == 1)   ? 8192
                        : (variant == 3) ? 32768
+                       : (variant == 4) ? 65536
                                         : 16384;
         uint32_t blk = (variant >= 2) ? 512 : 256;
         uint64_t windows = (total_out + win - 1) / win;
-        uint32_t grid = windows > 4096 ? 4096 : (uint32_t)windows;
+        uint64_t gcap = 4096;
+        if (const char* g = getenv("DBEEL_COPY_GRID")) {
+            long v = atol(g);
+            if (v >= 64) gcap = (uint64_t)v;
+        }
+        uint32_t grid = windows > gcap ? (uint32_t)gcap : (uint32_t)windows;
         hipLaunchKernelGGL(k_winmap, dim3(pick_grid(windows, 256)), dim3(256),
                            0, s, job->d_outindex, n_surv, total_out, win,
                            job->d_winp0);
@@ -1694,6 +1700,7 @@ extern "C" int dbeel_gpu_job_run(dbeel_gpu_job* job, int keep_tombstones,
         if (variant == 1) kc = k_copy<256, 8192>;
         if (variant == 2) kc = k_copy<512, 16384>;
         if (variant == 3) kc = k_copy<512, 32768>;
+        if (variant == 4) kc = k_copy<512, 65536>;
         hipLaunchKernelGGL(kc, dim3(grid), dim3(blk), 0, s,
                            job->d_outindex, job->d_srcmap, job->d_winp0,
                            n_surv, total_out, job->d_outdata);

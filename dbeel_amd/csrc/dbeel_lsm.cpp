@@ -311,6 +311,23 @@ extern "C" int dbeel_lsm_compact(const char* dir_c, const uint64_t* indices,
     return DBEEL_OK;
 }
 
+extern "C" int dbeel_lsm_major_compact(const char* dir_c, int device,
+                                       uint64_t sstable_bloom_min_size,
+                                       uint64_t* out_entries_written) {
+    if (!dir_c) return DBEEL_ERR_INVALID_ARG;
+    std::string dir(dir_c);
+    std::vector<uint64_t> idxs = scan_indices(dir, "index");
+    if (out_entries_written) *out_entries_written = 0;
+    if (idxs.size() < 2) return DBEEL_OK;
+    uint64_t out_index = 1;
+    for (uint64_t i : idxs)
+        if (i % 2 == 1 && i + 2 > out_index) out_index = i + 2;
+    /* the group covers every live sstable -> dropping tombstones is safe */
+    return dbeel_lsm_compact(dir_c, idxs.data(), idxs.size(), out_index,
+                             /*keep_tombstones=*/0, device,
+                             sstable_bloom_min_size, out_entries_written);
+}
+
 extern "C" int dbeel_lsm_compact_tree(const char* dir_c,
                                       uint64_t compaction_factor, int device,
                                       uint64_t sstable_bloom_min_size,

@@ -82,6 +82,26 @@ def compact_tree(dir: str, compaction_factor: int = 2, device: int = 0,
     return int(n.value)
 
 
+def major_compact(dir: str, device: int = 0,
+                  bloom_min_size: int = DEFAULT_BLOOM_MIN_SIZE) -> int:
+    """Merge EVERY live sstable into one run, dropping tombstones (safe:
+    the group covers everything). Bounds tombstone buildup under the
+    conservative full-coverage rule (include/dbeel_lsm.h)."""
+    lib = load()
+    if not hasattr(lib, "_major_ready"):
+        lib.dbeel_lsm_major_compact.restype = ctypes.c_int
+        lib.dbeel_lsm_major_compact.argtypes = [
+            ctypes.c_char_p, ctypes.c_int, ctypes.c_uint64,
+            ctypes.POINTER(ctypes.c_uint64),
+        ]
+        lib._major_ready = True
+    n = ctypes.c_uint64()
+    rc = lib.dbeel_lsm_major_compact(dir.encode(), device, bloom_min_size,
+                                     ctypes.byref(n))
+    _check(rc, lib)
+    return int(n.value)
+
+
 def bloom_contains(bloom_bytes: bytes, key: bytes) -> bool:
     lib = load()
     bb = (ctypes.c_uint8 * len(bloom_bytes)).from_buffer_copy(bloom_bytes)

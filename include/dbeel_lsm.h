@@ -60,6 +60,18 @@ int dbeel_lsm_compact_tree(const char* dir, uint64_t compaction_factor,
                            int device, uint64_t sstable_bloom_min_size,
                            uint32_t* out_n_compactions);
 
+/* Major compaction: merge EVERY live sstable in dir into one run at the
+ * next odd output index, dropping tombstones (safe: nothing outside the
+ * group can resurrect a deleted key). Bounds the tombstone buildup that
+ * the conservative full-coverage rule in dbeel_lsm_compact_tree can
+ * otherwise let accumulate (a steady-state tree may never form a
+ * full-coverage group on its own — ADVICE r01); call periodically or
+ * when space amplification matters. No-op (0 compactions) when fewer
+ * than 2 sstables exist. */
+int dbeel_lsm_major_compact(const char* dir, int device,
+                            uint64_t sstable_bloom_min_size,
+                            uint64_t* out_entries_written);
+
 /* Membership test against an engine-format bloom file's bytes.
  * *out = 1 if possibly present, 0 if definitely absent. */
 int dbeel_bloom_contains(const uint8_t* bloom_bytes, size_t bloom_len,

@@ -30,6 +30,7 @@ LSM_SYMBOLS = [
     "dbeel_lsm_compact",
     "dbeel_lsm_replay",
     "dbeel_lsm_compact_tree",
+    "dbeel_lsm_major_compact",
     "dbeel_bloom_contains",
 ]
 

@@ -52,6 +52,29 @@ def test_compact_files_end_to_end(tmp_path):
     assert fp < 60  # ~1% expected, generous bound
 
 
+def test_major_compact_bounds_tombstones(tmp_path):
+    """dbeel_lsm_major_compact merges every live sstable and drops
+    tombstones (safe: full coverage) — the operator tool bounding the
+    buildup the conservative compact_tree rule can leave."""
+    d = str(tmp_path)
+    runs = make_runs(4, 3000, 16, 64, overlap_frac=0.5, tombstone_frac=0.3,
+                     seed=7)
+    _write_runs(d, [0, 2, 4, 6], runs)
+    n = lsm.major_compact(d, device=0, bloom_min_size=1 << 30)
+    od, oi, on = oracle.compact(runs, keep_tombstones=False)
+    assert n == on
+    # single surviving run at the next odd index, tombstones gone
+    import glob
+
+    left = sorted(glob.glob(f"{d}/*.index"))
+    assert len(left) == 1 and left[0].endswith(f"{7:020d}.index")
+    data, index = lsm.read_run_files(d, 7)
+    assert data == od and index == oi
+    assert all(not e.is_tombstone for e in parse_run(data, index))
+    # idempotent-ish: a second call with one sstable is a no-op
+    assert lsm.major_compact(d, device=0) == 0
+
+
 def test_compact_files_no_bloom_below_threshold(tmp_path):
     d = str(tmp_path)
     runs = make_runs(2, 200, 16, 32, seed=7)

@@ -20,6 +20,7 @@ VARIANTS = {
     1: "256thr x 8KiB",
     2: "512thr x 16KiB",
     3: "512thr x 32KiB",
+    4: "512thr x 64KiB",
 }
 
 
@@ -27,24 +28,44 @@ def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--rounds", type=int, default=12)
     ap.add_argument("--scale", type=float, default=1.0)
+    ap.add_argument("--workload", default="cfg3", choices=["cfg3", "cfg5"])
+    ap.add_argument("--grids", default="",
+                    help="comma list of DBEEL_COPY_GRID caps to sweep "
+                         "(per variant); empty = default 4096")
     args = ap.parse_args()
 
-    runs = make_config("cfg3", scale=args.scale)
+    if args.workload == "cfg5":
+        from dbeel_amd.genruns import CONFIGS, make_runs_varkey
+
+        cfg = dict(CONFIGS["cfg5"])
+        if args.scale != 1.0:
+            cfg["entries_per_run"] = int(cfg["entries_per_run"] * args.scale)
+        runs = make_runs_varkey(seed=0xDBEE1, **cfg)
+    else:
+        runs = make_config("cfg3", scale=args.scale)
     job = dbeel_amd.Job(runs, device=0)
     # warmup
     for v in VARIANTS:
         os.environ["DBEEL_COPY_VARIANT"] = str(v)
         job.run(False)
 
-    times = {v: [] for v in VARIANTS}
+    grids = [int(g) for g in args.grids.split(",") if g] or [None]
+    combos = [(v, g) for v in VARIANTS for g in grids]
+    times = {c: [] for c in combos}
     for r in range(args.rounds):
-        for v in VARIANTS:
+        for v, g in combos:
             os.environ["DBEEL_COPY_VARIANT"] = str(v)
+            if g is None:
+                os.environ.pop("DBEEL_COPY_GRID", None)
+            else:
+                os.environ["DBEEL_COPY_GRID"] = str(g)
             _, _, t = job.run(False)
-            times[v].append(t["copy_ms"])
-    for v, name in VARIANTS.items():
-        ts = sorted(times[v])
-        print(f"variant {v} ({name}): median={statistics.median(ts):.3f} ms "
+            times[(v, g)].append(t["copy_ms"])
+    for (v, g), ts in times.items():
+        ts = sorted(ts)
+        gs = f" grid={g}" if g else ""
+        print(f"variant {v} ({VARIANTS[v]}){gs}: "
+              f"median={statistics.median(ts):.3f} ms "
               f"min={ts[0]:.3f} max={ts[-1]:.3f}")
     job.close()
 
