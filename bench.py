@@ -111,9 +111,9 @@ def make_workload(rank: int, scale: float, workload: str = "cfg3"):
 
 
 def aux_record_bytes(key_size: int) -> int:
-    # aux tier (dbeel_gpu.hip AuxT): 16-B records for klen<=20, 32-B for
-    # klen<=36, 64-B beyond
-    return 16 if key_size <= 20 else (32 if key_size <= 36 else 64)
+    # aux tier (dbeel_gpu.hip AuxT): 16-B ts-less records for klen<=20,
+    # 48-B with staged ts for klen<=32, 64-B with ts beyond
+    return 16 if key_size <= 20 else (48 if key_size <= 32 else 64)
 
 
 def main():

@@ -213,13 +213,25 @@ __device__ __forceinline__ uint64_t key_prefix(const uint8_t* key,
  * indexes; longer keys stay CORRECT in any tier (blob fallback when two
  * keys tie through 8+KB bytes), smaller tiers are purely a traffic
  * optimization. */
-template <int KB> struct AuxT {
+template <int KB, bool TS> struct AuxT;
+template <int KB> struct AuxT<KB, false> {
     uint8_t key[KB]; /* key bytes 8..8+KB, zero-padded */
     uint32_t klen;
 };
-static_assert(sizeof(AuxT<12>) == 16, "tier-0 aux must be 16B");
-static_assert(sizeof(AuxT<28>) == 32, "tier-1 aux must be 32B");
-static_assert(sizeof(AuxT<60>) == 64, "tier-2 aux must be 64B");
+template <int KB> struct AuxT<KB, true> {
+    uint8_t key[KB];
+    uint32_t klen;
+    uint32_t rsv;
+    uint64_t ts_lo; /* timestamp i128 LE halves — staged for tiers where
+                       cross-run duplicates are expected so the (ts, run)
+                       tie-break stays dense instead of 2 scattered blob
+                       lines per comparison (PMC r02: blob ts reads cost
+                       corank +1.3 GB on cfg3) */
+    int64_t ts_hi;
+};
+static_assert(sizeof(AuxT<12, false>) == 16, "tier-0 aux must be 16B");
+static_assert(sizeof(AuxT<24, true>) == 48, "tier-1 aux must be 48B");
+static_assert(sizeof(AuxT<40, true>) == 64, "tier-2 aux must be 64B");
 
 /* Key-only compare of entries A=(rA,iA), B=(rB,iB) via suffix records.
  * PRECONDITION: pfx(A) == pfx(B) (every caller compares pfx first), so
@@ -227,13 +239,13 @@ static_assert(sizeof(AuxT<60>) == 64, "tier-2 aux must be 64B");
  * is consistent: equality there with different klen means one key is a
  * zero-extension of the other within 8 bytes, which the klen comparison
  * below orders correctly (a strict prefix sorts first). */
-template <int KB>
+template <int KB, bool TS>
 __device__ __forceinline__ int cmp_keys_sfx(const RunsDesc& R,
-                                            const AuxT<KB>* aux, int rA,
+                                            const AuxT<KB, TS>* aux, int rA,
                                             uint64_t iA, int rB,
                                             uint64_t iB) {
-    const AuxT<KB>* a = aux + R.entry_base[rA] + iA;
-    const AuxT<KB>* b = aux + R.entry_base[rB] + iB;
+    const AuxT<KB, TS>* a = aux + R.entry_base[rA] + iA;
+    const AuxT<KB, TS>* b = aux + R.entry_base[rB] + iB;
     uint32_t la = a->klen, lb = b->klen;
     uint32_t n = la < lb ? la : lb;
     uint32_t ns = n > 8 ? n - 8 : 0; /* suffix bytes to compare */
@@ -285,23 +297,31 @@ __device__ __forceinline__ int cmp_ts_run_blob(const RunsDesc& R, int rA,
 }
 
 /* Full-order compare (key, timestamp, run index — lsm_tree.rs:52-71). */
-template <int KB>
+template <int KB, bool TS>
 __device__ __forceinline__ int cmp_sfx_full(const RunsDesc& R,
-                                            const AuxT<KB>* aux, int rA,
+                                            const AuxT<KB, TS>* aux, int rA,
                                             uint64_t iA, int rB,
                                             uint64_t iB) {
-    int c = cmp_keys_sfx<KB>(R, aux, rA, iA, rB, iB);
+    int c = cmp_keys_sfx<KB, TS>(R, aux, rA, iA, rB, iB);
     if (c) return c;
-    return cmp_ts_run_blob(R, rA, iA, rB, iB);
+    if constexpr (TS) {
+        const AuxT<KB, TS>* a = aux + R.entry_base[rA] + iA;
+        const AuxT<KB, TS>* b = aux + R.entry_base[rB] + iB;
+        if (a->ts_hi != b->ts_hi) return a->ts_hi < b->ts_hi ? -1 : 1;
+        if (a->ts_lo != b->ts_lo) return a->ts_lo < b->ts_lo ? -1 : 1;
+        return rA < rB ? -1 : (rA > rB ? 1 : 0);
+    } else {
+        return cmp_ts_run_blob(R, rA, iA, rB, iB);
+    }
 }
 
 /* Validates every entry (bounds + bincode field cross-check,
  * read_next_entry lsm_tree.rs:1158-70) and extracts the dense key-prefix
  * and aux arrays the merge runs on. Run sortedness is checked in
  * k_rankreduce (on the dense pfx/aux arrays). */
-template <int KB>
+template <int KB, bool TS>
 __global__ void k_prepare(RunsDesc R, uint64_t g0, uint64_t g1,
-                          uint64_t* pfx, AuxT<KB>* aux, uint32_t* err) {
+                          uint64_t* pfx, AuxT<KB, TS>* aux, uint32_t* err) {
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
     for (uint64_t g = g0 + (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
          g < g1; g += stride) {
@@ -313,13 +333,17 @@ __global__ void k_prepare(RunsDesc R, uint64_t g0, uint64_t g1,
         if (!load_entry(R, r, i, e)) {
             atomicOr(err, DERR_CORRUPT);
             pfx[g] = 0;
-            AuxT<KB> z = {};
+            AuxT<KB, TS> z = {};
             aux[g] = z;
             continue;
         }
         pfx[g] = key_prefix(e.key, e.klen);
-        AuxT<KB> a;
+        AuxT<KB, TS> a;
         a.klen = (uint32_t)e.klen;
+        if constexpr (TS) {
+            a.rsv = 0;
+            load_ts(e, a.ts_lo, a.ts_hi);
+        }
         /* stage key SUFFIX bytes 8..8+KB, zero-padded. 8-byte chunked
          * copy; reading up to 7 bytes past the key is safe (the 8-byte
          * data_len field follows it inside the entry) and the over-read
@@ -386,9 +410,9 @@ struct PairDesc {
     uint64_t chunk_base; /* exclusive prefix sum of per-pair windows */
 };
 
-template <int KB>
+template <int KB, bool TS>
 __global__ __launch_bounds__(CORANK_BLOCK) void k_corank(
-    RunsDesc R, const uint64_t* pfx, const AuxT<KB>* aux,
+    RunsDesc R, const uint64_t* pfx, const AuxT<KB, TS>* aux,
     const PairDesc* pairs, uint32_t n_pairs, uint64_t total_chunks,
     uint32_t* cr) {
     __shared__ uint64_t s_pfx[CORANK_BLOCK_POS + 2];
@@ -428,8 +452,8 @@ __global__ __launch_bounds__(CORANK_BLOCK) void k_corank(
                 uint64_t pa = gpa[mid], pb = gpb[diag - mid - 1];
                 int c = (pa != pb)
                             ? (pa < pb ? -1 : 1)
-                            : cmp_sfx_full<KB>(R, aux, a, mid, b,
-                                               diag - mid - 1);
+                            : cmp_sfx_full<KB, TS>(R, aux, a, mid, b,
+                                                   diag - mid - 1);
                 if (c < 0)
                     slo = mid + 1;
                 else
@@ -464,8 +488,8 @@ __global__ __launch_bounds__(CORANK_BLOCK) void k_corank(
                 uint32_t mid = (slo + shi) >> 1;
                 uint64_t pa = sA[mid], pb = sB[d - mid - 1];
                 int c = (pa != pb) ? (pa < pb ? -1 : 1)
-                                   : cmp_sfx_full<KB>(R, aux, a, iaS + mid,
-                                                      b, ibS + d - mid - 1);
+                                   : cmp_sfx_full<KB, TS>(R, aux, a, iaS + mid,
+                                                          b, ibS + d - mid - 1);
                 if (c < 0)
                     slo = mid + 1;
                 else
@@ -485,8 +509,8 @@ __global__ __launch_bounds__(CORANK_BLOCK) void k_corank(
                     uint64_t pa = sA[ja], pb = sB[jb];
                     take_a = (pa != pb)
                                  ? (pa < pb)
-                                 : (cmp_sfx_full<KB>(R, aux, a, iaS + ja, b,
-                                                     ibS + jb) < 0);
+                                 : (cmp_sfx_full<KB, TS>(R, aux, a, iaS + ja, b,
+                                                         ibS + jb) < 0);
                 }
                 if (take_a) {
                     uint64_t gib = ibS + jb;
@@ -496,8 +520,8 @@ __global__ __launch_bounds__(CORANK_BLOCK) void k_corank(
                         uint64_t pb =
                             (jb < lenB) ? sB[jb] : gpb[gib];
                         if (pb == sA[ja] &&
-                            cmp_keys_sfx<KB>(R, aux, a, iaS + ja, b,
-                                             gib) == 0)
+                            cmp_keys_sfx<KB, TS>(R, aux, a, iaS + ja, b,
+                                                 gib) == 0)
                             v |= CR_LOSER;
                     }
                     s_cr[ja] = v;
@@ -509,8 +533,8 @@ __global__ __launch_bounds__(CORANK_BLOCK) void k_corank(
                         uint64_t pa =
                             (ja < lenA) ? sA[ja] : gpa[gia];
                         if (pa == sB[jb] &&
-                            cmp_keys_sfx<KB>(R, aux, b, ibS + jb, a,
-                                             gia) == 0)
+                            cmp_keys_sfx<KB, TS>(R, aux, b, ibS + jb, a,
+                                                 gia) == 0)
                             v |= CR_LOSER;
                     }
                     s_cr[lenA + jb] = v;
@@ -552,9 +576,9 @@ static_assert(sizeof(RankRec) == 16, "rank record must be 16B");
  * tombstone rules, and emit the rank-indexed scratch records. Also checks
  * each run is strictly sorted by key (flush invariant,
  * lsm_tree.rs:925-946) on the dense pfx/aux arrays. */
-template <int KB>
+template <int KB, bool TS>
 __global__ void k_rankreduce(RunsDesc R, const uint64_t* pfx,
-                             const AuxT<KB>* aux, const uint32_t* cr,
+                             const AuxT<KB, TS>* aux, const uint32_t* cr,
                              RankRec* rrec, int keep_tombstones,
                              uint32_t* err) {
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
@@ -575,7 +599,7 @@ __global__ void k_rankreduce(RunsDesc R, const uint64_t* pfx,
             uint64_t p0 = pfx[g], p1 = pfx[g + 1];
             if (p0 > p1 ||
                 (p0 == p1 &&
-                 cmp_keys_sfx<KB>(R, aux, r, i, r, i + 1) >= 0))
+                 cmp_keys_sfx<KB, TS>(R, aux, r, i, r, i + 1) >= 0))
                 atomicOr(err, DERR_UNSORTED);
         }
 
@@ -1322,11 +1346,11 @@ extern "C" int dbeel_gpu_job_create(const dbeel_run_view* runs, size_t n_runs,
                                  hipMemcpyDeviceToHost, job->stream));
         JOB_CHECK(hipStreamSynchronize(job->stream));
         uint32_t maxklen = maxks > 8 ? maxks - 8 : 0;
-        job->aux_kind = maxklen <= 20 ? 0 : (maxklen <= 36 ? 1 : 2);
+        job->aux_kind = maxklen <= 20 ? 0 : (maxklen <= 32 ? 1 : 2);
     } else {
         job->aux_kind = 0;
     }
-    uint64_t aux_sz = job->aux_kind == 0 ? 16 : (job->aux_kind == 1 ? 32
+    uint64_t aux_sz = job->aux_kind == 0 ? 16 : (job->aux_kind == 1 ? 48
                                                                     : 64);
     JOB_CHECK(hipMalloc(&job->d_aux, n * aux_sz));
 
@@ -1520,15 +1544,15 @@ extern "C" int dbeel_gpu_job_ingest(dbeel_gpu_job* job,
                 uint64_t g1 = job->desc.entry_base[r] + e1;
                 uint32_t grid = pick_grid(g1 - g0, 256);
                 switch (job->aux_kind) {
-#define PREP_RANGE(KB)                                                      \
-    case (KB == 12 ? 0 : (KB == 28 ? 1 : 2)):                               \
-        hipLaunchKernelGGL(k_prepare<KB>, dim3(grid), dim3(256), 0, ks,     \
-                           job->desc, g0, g1, job->d_pfx,                   \
-                           (AuxT<KB>*)job->d_aux, job->d_err);              \
+#define PREP_RANGE(KB, TS)                                                  \
+    case (KB == 12 ? 0 : (KB == 24 ? 1 : 2)):                               \
+        hipLaunchKernelGGL((k_prepare<KB, TS>), dim3(grid), dim3(256), 0,   \
+                           ks, job->desc, g0, g1, job->d_pfx,               \
+                           (AuxT<KB, TS>*)job->d_aux, job->d_err);          \
         break;
-                    PREP_RANGE(12)
-                    PREP_RANGE(28)
-                    PREP_RANGE(60)
+                    PREP_RANGE(12, false)
+                    PREP_RANGE(24, true)
+                    PREP_RANGE(40, true)
 #undef PREP_RANGE
                 }
             }
@@ -1602,27 +1626,27 @@ extern "C" int dbeel_gpu_job_run(dbeel_gpu_job* job, int keep_tombstones,
         uint64_t cgrid = job->total_chunks;
         if (cgrid > 16384) cgrid = 16384;
         switch (job->aux_kind) {
-#define STAGE1(KB)                                                          \
-    case (KB == 12 ? 0 : (KB == 28 ? 1 : 2)):                               \
+#define STAGE1(KB, TS)                                                      \
+    case (KB == 12 ? 0 : (KB == 24 ? 1 : 2)):                               \
         if (!skip_prep)                                                     \
-            hipLaunchKernelGGL(k_prepare<KB>, dim3(grid), dim3(256), 0, s,  \
-                               job->desc, (uint64_t)0, n, job->d_pfx,       \
-                               (AuxT<KB>*)job->d_aux, job->d_err);          \
+            hipLaunchKernelGGL((k_prepare<KB, TS>), dim3(grid), dim3(256),  \
+                               0, s, job->desc, (uint64_t)0, n, job->d_pfx, \
+                               (AuxT<KB, TS>*)job->d_aux, job->d_err);      \
         hipEventRecord(job->ev[6], s);                                      \
         if (job->n_pairs)                                                   \
-            hipLaunchKernelGGL(k_corank<KB>, dim3((uint32_t)cgrid),         \
+            hipLaunchKernelGGL((k_corank<KB, TS>), dim3((uint32_t)cgrid),   \
                                dim3(CORANK_BLOCK), 0, s, job->desc,         \
-                               job->d_pfx, (const AuxT<KB>*)job->d_aux,     \
+                               job->d_pfx, (const AuxT<KB, TS>*)job->d_aux, \
                                (const PairDesc*)job->d_pairs, job->n_pairs, \
                                job->total_chunks, job->d_cr);               \
-        hipLaunchKernelGGL(k_rankreduce<KB>, dim3(grid), dim3(256), 0, s,   \
-                           job->desc, job->d_pfx,                           \
-                           (const AuxT<KB>*)job->d_aux, job->d_cr,          \
+        hipLaunchKernelGGL((k_rankreduce<KB, TS>), dim3(grid), dim3(256),   \
+                           0, s, job->desc, job->d_pfx,                     \
+                           (const AuxT<KB, TS>*)job->d_aux, job->d_cr,      \
                            job->d_rank, keep_tombstones, job->d_err);       \
         break;
-            STAGE1(12)
-            STAGE1(28)
-            STAGE1(60)
+            STAGE1(12, false)
+            STAGE1(24, true)
+            STAGE1(40, true)
 #undef STAGE1
         }
     } else {
