@@ -1603,6 +1603,55 @@ static uint32_t pick_grid(uint64_t work_items, uint32_t block) {
     return (uint32_t)blocks;
 }
 
+
+/* Copy-geometry selection measured by interleaved A/B (tools/ab_copy.py,
+ * r02): ~1 KiB survivors (cfg3) run 11% faster with 64 KiB windows /
+ * 512 threads (fewer window transitions+barriers per block); 4 KiB
+ * survivors (cfg5) want MORE blocks (grid 16384) at the default window
+ * (the 4096 cap left most waves parked); sub-512-B survivors keep the
+ * default. Env DBEEL_COPY_VARIANT / DBEEL_COPY_GRID override for A/Bs. */
+static void launch_copy(hipStream_t s, const uint8_t* out_index,
+                        const uint64_t* src_map, uint32_t* win_p0,
+                        uint64_t n_surv, uint64_t total_out,
+                        uint8_t* out_data) {
+    int variant = -1;
+    if (const char* v = getenv("DBEEL_COPY_VARIANT")) variant = atoi(v);
+    uint64_t gcap = 0;
+    if (const char* g = getenv("DBEEL_COPY_GRID")) {
+        long v = atol(g);
+        if (v >= 64) gcap = (uint64_t)v;
+    }
+    if (variant < 0) {
+        uint64_t avg = n_surv ? total_out / n_surv : 0;
+        if (avg > 2048) {
+            variant = 0;
+            if (!gcap) gcap = 16384;
+        } else if (avg >= 512) {
+            variant = 4;
+        } else {
+            variant = 0;
+        }
+    }
+    if (!gcap) gcap = 4096;
+    uint32_t win = (variant == 1)   ? 8192
+                   : (variant == 3) ? 32768
+                   : (variant == 4) ? 65536
+                                    : 16384;
+    uint32_t blk = (variant >= 2) ? 512 : 256;
+    uint64_t windows = (total_out + win - 1) / win;
+    uint32_t grid = windows > gcap ? (uint32_t)gcap : (uint32_t)windows;
+    hipLaunchKernelGGL(k_winmap, dim3(pick_grid(windows, 256)), dim3(256),
+                       0, s, out_index, n_surv, total_out, win, win_p0);
+    void (*kc)(const uint8_t*, const uint64_t*, const uint32_t*, uint64_t,
+               uint64_t, uint8_t*) = k_copy<256, 16384>;
+    if (variant == 1) kc = k_copy<256, 8192>;
+    if (variant == 2) kc = k_copy<512, 16384>;
+    if (variant == 3) kc = k_copy<512, 32768>;
+    if (variant == 4) kc = k_copy<512, 65536>;
+    hipLaunchKernelGGL(kc, dim3(grid), dim3(blk), 0, s, out_index, src_map,
+                       win_p0, n_surv, total_out, out_data);
+}
+
 extern "C" int dbeel_gpu_job_run(dbeel_gpu_job* job, int keep_tombstones,
                                  uint64_t* out_data_len,
                                  uint64_t* out_entries,
@@ -1701,34 +1750,9 @@ extern "C" int dbeel_gpu_job_run(dbeel_gpu_job* job, int keep_tombstones,
     uint64_t n_surv = (uint64_t)last_pos + last_kept;
 
     HIP_CHECK(hipEventRecord(job->ev[4], s));
-    if (total_out) {
-        int variant = 0;
-        if (const char* v = getenv("DBEEL_COPY_VARIANT")) variant = atoi(v);
-        uint32_t win = (variant == 1)   ? 8192
-                       : (variant == 3) ? 32768
-                       : (variant == 4) ? 65536
-                                        : 16384;
-        uint32_t blk = (variant >= 2) ? 512 : 256;
-        uint64_t windows = (total_out + win - 1) / win;
-        uint64_t gcap = 4096;
-        if (const char* g = getenv("DBEEL_COPY_GRID")) {
-            long v = atol(g);
-            if (v >= 64) gcap = (uint64_t)v;
-        }
-        uint32_t grid = windows > gcap ? (uint32_t)gcap : (uint32_t)windows;
-        hipLaunchKernelGGL(k_winmap, dim3(pick_grid(windows, 256)), dim3(256),
-                           0, s, job->d_outindex, n_surv, total_out, win,
-                           job->d_winp0);
-        void (*kc)(const uint8_t*, const uint64_t*, const uint32_t*,
-                   uint64_t, uint64_t, uint8_t*) = k_copy<256, 16384>;
-        if (variant == 1) kc = k_copy<256, 8192>;
-        if (variant == 2) kc = k_copy<512, 16384>;
-        if (variant == 3) kc = k_copy<512, 32768>;
-        if (variant == 4) kc = k_copy<512, 65536>;
-        hipLaunchKernelGGL(kc, dim3(grid), dim3(blk), 0, s,
-                           job->d_outindex, job->d_srcmap, job->d_winp0,
-                           n_surv, total_out, job->d_outdata);
-    }
+    if (total_out)
+        launch_copy(s, job->d_outindex, job->d_srcmap, job->d_winp0,
+                    n_surv, total_out, job->d_outdata);
     HIP_CHECK(hipEventRecord(job->ev[5], s));
     HIP_CHECK(hipStreamSynchronize(s));
     HIP_CHECK(hipGetLastError());
@@ -2213,18 +2237,9 @@ extern "C" int dbeel_gpu_scan(const dbeel_run_view* runs, size_t n_runs,
         uint64_t total_out =
             last_off + (last_kept ? last_rec.full_size : 0);
         uint64_t n_surv = (uint64_t)last_pos + last_kept;
-        if (total_out) {
-            uint32_t win = 16384;
-            uint64_t windows = (total_out + win - 1) / win;
-            uint32_t grid = windows > 4096 ? 4096 : (uint32_t)windows;
-            hipLaunchKernelGGL(k_winmap, dim3(pick_grid(windows, 256)),
-                               dim3(256), 0, s, job->d_outindex, n_surv,
-                               total_out, win, job->d_winp0);
-            hipLaunchKernelGGL((k_copy<256, 16384>), dim3(grid), dim3(256),
-                               0, s, job->d_outindex, job->d_srcmap,
-                               job->d_winp0, n_surv, total_out,
-                               job->d_outdata);
-        }
+        if (total_out)
+            launch_copy(s, job->d_outindex, job->d_srcmap, job->d_winp0,
+                        n_surv, total_out, job->d_outdata);
         e = hipStreamSynchronize(s);
         if (e == hipSuccess) e = hipGetLastError();
         if (e != hipSuccess) {

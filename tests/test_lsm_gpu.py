@@ -63,12 +63,13 @@ def test_major_compact_bounds_tombstones(tmp_path):
     n = lsm.major_compact(d, device=0, bloom_min_size=1 << 30)
     od, oi, on = oracle.compact(runs, keep_tombstones=False)
     assert n == on
-    # single surviving run at the next odd index, tombstones gone
+    # single surviving run at the next odd index (no odd sstables existed
+    # -> index 1, tasks/compaction.rs:38-43), tombstones gone
     import glob
 
     left = sorted(glob.glob(f"{d}/*.index"))
-    assert len(left) == 1 and left[0].endswith(f"{7:020d}.index")
-    data, index = lsm.read_run_files(d, 7)
+    assert len(left) == 1 and left[0].endswith(f"{1:020d}.index"), left
+    data, index = lsm.read_run_files(d, 1)
     assert data == od and index == oi
     assert all(not e.is_tombstone for e in parse_run(data, index))
     # idempotent-ish: a second call with one sstable is a no-op
