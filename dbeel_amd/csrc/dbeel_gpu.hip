@@ -86,6 +86,8 @@ struct RunsDesc {
     uint64_t data_len[MAX_RUNS];
     uint64_t count[MAX_RUNS];       /* entries per run                  */
     uint64_t entry_base[MAX_RUNS];  /* exclusive prefix sum of count    */
+    uint8_t job_lo[MAX_RUNS];       /* run-range [lo, hi) of the        */
+    uint8_t job_hi[MAX_RUNS];       /* INDEPENDENT job owning run r     */
     int n_runs;
     uint64_t total;                 /* sum of count                     */
 };
@@ -543,10 +545,14 @@ __global__ __launch_bounds__(CORANK_BLOCK) void k_corank(
             }
         }
         __syncthreads();
-        /* coalesced writeback of both contiguous output ranges */
+        /* coalesced writeback of both contiguous output ranges.
+         * Crossrank slots are JOB-local (a batched job's runs only pair
+         * within their job) */
         {
-            uint32_t sa = (uint32_t)(b < a ? b : b - 1); /* slot of b in a */
-            uint32_t sb = (uint32_t)(a < b ? a : a - 1); /* slot of a in b */
+            uint32_t al = (uint32_t)a - R.job_lo[a];
+            uint32_t bl = (uint32_t)b - R.job_lo[b];
+            uint32_t sa = bl < al ? bl : bl - 1; /* slot of b in a */
+            uint32_t sb = al < bl ? al : al - 1; /* slot of a in b */
             uint32_t* cra = cr + (uint64_t)sa * R.total + R.entry_base[a] +
                             iaS;
             uint32_t* crb = cr + (uint64_t)sb * R.total + R.entry_base[b] +
@@ -603,9 +609,13 @@ __global__ void k_rankreduce(RunsDesc R, const uint64_t* pfx,
                 atomicOr(err, DERR_UNSORTED);
         }
 
-        uint64_t rank = i;
+        /* rank is local to the entry's JOB (independent jobs share one
+         * launch); the job's entry base turns it into the global rrec
+         * slot, so job outputs concatenate in job order */
+        uint64_t rank = i + R.entry_base[R.job_lo[r]];
         bool winner = true;
-        for (int s = 0; s + 1 < R.n_runs; s++) {
+        int nslots = (int)R.job_hi[r] - (int)R.job_lo[r] - 1;
+        for (int s = 0; s < nslots; s++) {
             uint32_t v = cr[(uint64_t)s * R.total + g];
             rank += v & CR_MASK;
             winner &= !(v & CR_LOSER);
@@ -1129,6 +1139,7 @@ struct dbeel_gpu_job {
     uint64_t total_entries = 0;
     uint64_t total_data_bytes = 0;
     uint64_t input_bytes = 0;
+    std::vector<uint64_t> job_gbase; /* entry base per job (+ total) */
     /* last-run results */
     uint64_t out_data_len = 0;
     uint64_t out_entries = 0;
@@ -1178,8 +1189,9 @@ static int validate_runs(const dbeel_run_view* runs, size_t n_runs) {
     return DBEEL_OK;
 }
 
-extern "C" int dbeel_gpu_job_create(const dbeel_run_view* runs, size_t n_runs,
-                                    int device, dbeel_gpu_job** out_job) {
+static int job_create_impl(const dbeel_run_view* runs, size_t n_runs,
+                           const uint32_t* runs_per_job, size_t n_jobs,
+                           int device, dbeel_gpu_job** out_job) {
     g_err[0] = 0;
     if (!out_job) {
         set_err("out_job is null");
@@ -1188,6 +1200,21 @@ extern "C" int dbeel_gpu_job_create(const dbeel_run_view* runs, size_t n_runs,
     *out_job = nullptr;
     int rc = validate_runs(runs, n_runs);
     if (rc) return rc;
+    {
+        uint64_t sum = 0;
+        for (size_t j = 0; j < n_jobs; j++) {
+            if (!runs_per_job[j]) {
+                set_err("job %zu has zero runs", j);
+                return DBEEL_ERR_INVALID_ARG;
+            }
+            sum += runs_per_job[j];
+        }
+        if (sum != n_runs) {
+            set_err("runs_per_job sums to %llu != n_runs %zu",
+                    (unsigned long long)sum, n_runs);
+            return DBEEL_ERR_INVALID_ARG;
+        }
+    }
     if (device < 0) {
         set_err("device must be >= 0 (no CPU fallback in the product "
                 "library; the CPU restatement lives in oracle/liboracle.so "
@@ -1243,8 +1270,15 @@ extern "C" int dbeel_gpu_job_create(const dbeel_run_view* runs, size_t n_runs,
     JOB_CHECK(hipMalloc(&job->d_outdata, total_data ? total_data : 16));
     JOB_CHECK(hipMalloc(&job->d_err, 2 * sizeof(uint32_t)));
     JOB_CHECK(hipMalloc(&job->d_pfx, n * sizeof(uint64_t)));
+    uint32_t max_job_runs = 1;
+    {
+        for (size_t j = 0; j < n_jobs; j++)
+            if (runs_per_job[j] > max_job_runs)
+                max_job_runs = runs_per_job[j];
+    }
     JOB_CHECK(hipMalloc(&job->d_cr,
-                        (n_runs > 1 ? (n_runs - 1) * n : 1) * 4));
+                        (max_job_runs > 1 ? (max_job_runs - 1) * n : 1) *
+                            4));
     /* sized for the smallest copy-window variant (8 KiB) */
     JOB_CHECK(hipMalloc(&job->d_winp0,
                         (total_data / 8192 + 2) * sizeof(uint32_t)));
@@ -1270,6 +1304,16 @@ extern "C" int dbeel_gpu_job_create(const dbeel_run_view* runs, size_t n_runs,
     memset(&D, 0, sizeof D);
     D.n_runs = (int)n_runs;
     D.total = total;
+    {
+        uint32_t r0 = 0;
+        for (size_t j = 0; j < n_jobs; j++) {
+            for (uint32_t r = r0; r < r0 + runs_per_job[j]; r++) {
+                D.job_lo[r] = (uint8_t)r0;
+                D.job_hi[r] = (uint8_t)(r0 + runs_per_job[j]);
+            }
+            r0 += runs_per_job[j];
+        }
+    }
     uint64_t base = 0;
     for (size_t r = 0; r < n_runs; r++) {
         D.data[r] = p;
@@ -1301,7 +1345,7 @@ extern "C" int dbeel_gpu_job_create(const dbeel_run_view* runs, size_t n_runs,
         uint64_t cbase = 0;
         uint32_t np = 0;
         for (uint32_t a = 0; a < n_runs; a++)
-            for (uint32_t b = a + 1; b < n_runs; b++) {
+            for (uint32_t b = a + 1; b < D.job_hi[a]; b++) {
                 uint64_t len = D.count[a] + D.count[b];
                 if (!len) continue;
                 hp[np].a = a;
@@ -1327,6 +1371,14 @@ extern "C" int dbeel_gpu_job_create(const dbeel_run_view* runs, size_t n_runs,
         } else {
             free(hp);
         }
+    }
+    {
+        uint32_t r0 = 0;
+        for (size_t j = 0; j < n_jobs; j++) {
+            job->job_gbase.push_back(D.entry_base[r0]);
+            r0 += runs_per_job[j];
+        }
+        job->job_gbase.push_back(total);
     }
     JOB_CHECK(hipEventRecord(job->ev[1], job->stream));
     JOB_CHECK(hipStreamSynchronize(job->stream));
@@ -1357,6 +1409,94 @@ extern "C" int dbeel_gpu_job_create(const dbeel_run_view* runs, size_t n_runs,
     *out_job = job;
     return DBEEL_OK;
 #undef JOB_CHECK
+}
+
+extern "C" int dbeel_gpu_job_create(const dbeel_run_view* runs,
+                                    size_t n_runs, int device,
+                                    dbeel_gpu_job** out_job) {
+    uint32_t one[1] = {(uint32_t)n_runs};
+    return job_create_impl(runs, n_runs, one, 1, device, out_job);
+}
+
+/* Batched independent jobs in ONE launch set (BASELINE configs[3]'s
+ * 8-jobs-per-GPU shape): runs[] holds every job's runs back to back,
+ * runs_per_job[] their counts. Jobs never pair with each other (ranks,
+ * crossranks and winner flags are job-local); outputs land concatenated
+ * in job order and dbeel_gpu_job_fetch_job slices them back out. One
+ * kernel pipeline per step replaces 8 per-stream pipelines' launch
+ * storms and partial-fill tails. */
+extern "C" int dbeel_gpu_job_create_batched(const dbeel_run_view* runs,
+                                            size_t n_runs,
+                                            const uint32_t* runs_per_job,
+                                            size_t n_jobs, int device,
+                                            dbeel_gpu_job** out_job) {
+    if (!runs_per_job || !n_jobs) {
+        set_err("runs_per_job is null/empty");
+        return DBEEL_ERR_INVALID_ARG;
+    }
+    return job_create_impl(runs, n_runs, runs_per_job, n_jobs, device,
+                           out_job);
+}
+
+/* Per-job slice of a batched job's last result: data/index of job
+ * `job_idx`, index offsets rebased to the job's own run file. */
+extern "C" int dbeel_gpu_job_fetch_job(dbeel_gpu_job* job, size_t job_idx,
+                                       dbeel_compact_result* out) {
+    g_err[0] = 0;
+    if (!job || !out || !job->have_result ||
+        job_idx + 1 >= job->job_gbase.size()) {
+        set_err("fetch_job: no result / bad job index");
+        return DBEEL_ERR_INVALID_ARG;
+    }
+    HIP_CHECK(hipSetDevice(job->device));
+    memset(out, 0, sizeof *out);
+    uint64_t g0 = job->job_gbase[job_idx];
+    uint64_t g1 = job->job_gbase[job_idx + 1];
+    /* boundary offsets/positions: tiny D2H reads */
+    uint64_t off0 = 0, off1 = job->out_data_len;
+    uint32_t pos0 = 0, pos1 = (uint32_t)job->out_entries;
+    hipStream_t s = job->stream;
+    if (g0 > 0) {
+        HIP_CHECK(hipMemcpyAsync(&off0, job->d_dstoff + g0, 8,
+                                 hipMemcpyDeviceToHost, s));
+        HIP_CHECK(hipMemcpyAsync(&pos0, job->d_pos + g0, 4,
+                                 hipMemcpyDeviceToHost, s));
+    }
+    if (g1 < job->total_entries) {
+        HIP_CHECK(hipMemcpyAsync(&off1, job->d_dstoff + g1, 8,
+                                 hipMemcpyDeviceToHost, s));
+        HIP_CHECK(hipMemcpyAsync(&pos1, job->d_pos + g1, 4,
+                                 hipMemcpyDeviceToHost, s));
+    }
+    HIP_CHECK(hipStreamSynchronize(s));
+    uint64_t dlen = off1 - off0;
+    uint64_t nsurv = (uint64_t)pos1 - pos0;
+    out->data = (uint8_t*)malloc(dlen ? dlen : 1);
+    out->index = (uint8_t*)malloc(nsurv ? nsurv * 16 : 1);
+    if (!out->data || !out->index) {
+        free(out->data);
+        free(out->index);
+        memset(out, 0, sizeof *out);
+        set_err("host malloc failed");
+        return DBEEL_ERR_OOM;
+    }
+    if (dlen)
+        HIP_CHECK(hipMemcpyAsync(out->data, job->d_outdata + off0, dlen,
+                                 hipMemcpyDeviceToHost, s));
+    if (nsurv)
+        HIP_CHECK(hipMemcpyAsync(out->index,
+                                 job->d_outindex + (uint64_t)pos0 * 16,
+                                 nsurv * 16, hipMemcpyDeviceToHost, s));
+    HIP_CHECK(hipStreamSynchronize(s));
+    /* rebase offsets to the job's own file */
+    for (uint64_t i = 0; i < nsurv; i++) {
+        uint64_t o = ld_u64_host(out->index + i * 16) - off0;
+        memcpy(out->index + i * 16, &o, 8);
+    }
+    out->data_len = dlen;
+    out->index_len = nsurv * 16;
+    out->entries_written = nsurv;
+    return DBEEL_OK;
 }
 
 extern "C" void dbeel_gpu_job_destroy(dbeel_gpu_job* job) {

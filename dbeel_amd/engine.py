@@ -449,6 +449,55 @@ class Job:
             self._h = None
             self._keepalive = None
 
+
+class BatchJob(Job):
+    """Batched INDEPENDENT jobs in one launch set (one dbeel shard each,
+    BASELINE configs[3]): jobs = [runs, runs, ...]. run() executes every
+    job's merge in one kernel pipeline; fetch_job(j) returns job j's
+    (data, index, n) with offsets rebased to its own run file."""
+
+    def __init__(self, jobs, device: int = 0):
+        self._lib = load()
+        if not hasattr(self._lib, "_batch_ready"):
+            self._lib.dbeel_gpu_job_create_batched.restype = ctypes.c_int
+            self._lib.dbeel_gpu_job_create_batched.argtypes = [
+                ctypes.POINTER(RunView), ctypes.c_size_t,
+                ctypes.POINTER(ctypes.c_uint32), ctypes.c_size_t,
+                ctypes.c_int, ctypes.POINTER(ctypes.c_void_p),
+            ]
+            self._lib.dbeel_gpu_job_fetch_job.restype = ctypes.c_int
+            self._lib.dbeel_gpu_job_fetch_job.argtypes = [
+                ctypes.c_void_p, ctypes.c_size_t,
+                ctypes.POINTER(CompactResult),
+            ]
+            self._lib._batch_ready = True
+        flat = [rv for runs in jobs for rv in runs]
+        views, self._keepalive = _views(flat)
+        rpj = (ctypes.c_uint32 * len(jobs))(*[len(r) for r in jobs])
+        h = ctypes.c_void_p()
+        rc = self._lib.dbeel_gpu_job_create_batched(
+            views, len(flat), rpj, len(jobs), device, ctypes.byref(h)
+        )
+        if rc != 0:
+            raise DbeelGpuError(rc, self._lib.dbeel_gpu_last_error().decode())
+        self._h = h
+        self.n_jobs = len(jobs)
+        self.input_bytes = sum(v.data_len + v.index_len for v in views)
+
+    def fetch_job(self, j: int):
+        res = CompactResult()
+        rc = self._lib.dbeel_gpu_job_fetch_job(self._h, j,
+                                               ctypes.byref(res))
+        if rc != 0:
+            raise DbeelGpuError(rc, self._lib.dbeel_gpu_last_error().decode())
+        try:
+            data = _ptr_bytes(res.data, res.data_len)
+            index = _ptr_bytes(res.index, res.index_len)
+            n = int(res.entries_written)
+        finally:
+            self._lib.dbeel_gpu_result_free(ctypes.byref(res))
+        return data, index, n
+
     def __enter__(self):
         return self
 

@@ -339,6 +339,45 @@ def test_migration_scan_at_scale(engine):
     assert (gn, gi, gd) == (mn, mi, md)
 
 
+def test_batched_jobs_parity(engine):
+    """Batched independent jobs (one launch set, BASELINE configs[3]'s
+    8-jobs-per-GPU shape): each job's sliced-out result must be
+    bit-identical to compacting that job alone with the oracle — ranks,
+    crossrank slots and winner flags are job-local, and a key duplicated
+    ACROSS jobs must survive in every job (no cross-job dedup)."""
+    from dbeel_amd.engine import BatchJob
+
+    jobs = [
+        make_runs(4, 9_000, 16, 256, overlap_frac=0.4, tombstone_frac=0.1,
+                  seed=201),
+        make_runs(2, 5_000, 16, 64, overlap_frac=1.0, tombstone_frac=0.3,
+                  seed=202),
+        make_runs(1, 3_000, 16, 128, seed=203),   # single-run job
+        make_runs(8, 2_000, 32, 512, overlap_frac=0.5, tombstone_frac=0.05,
+                  seed=204),
+    ]
+    # duplicate one job entirely: same keys in two different jobs —
+    # cross-job isolation means both jobs keep their own winners
+    jobs.append(jobs[1])
+
+    for keep in (False, True):
+        with BatchJob(jobs, device=0) as bj:
+            out_bytes, out_entries, _ = bj.run(keep)
+            total_d = total_n = 0
+            for j, runs in enumerate(jobs):
+                gd, gi, gn = bj.fetch_job(j)
+                od, oi, on = oracle.compact(runs, keep_tombstones=keep)
+                assert gn == on, f"job {j}"
+                assert gi == oi, f"job {j}"
+                assert gd == od, f"job {j}"
+                total_d += len(gd)
+                total_n += gn
+            assert out_bytes == total_d
+            assert out_entries == total_n
+            # jobs 1 and 4 are identical inputs -> identical outputs
+            assert bj.fetch_job(1) == bj.fetch_job(4)
+
+
 def test_streamed_ingest_parity(engine):
     """Streamed pinned ingest (north_star: pinned host DRAM, chunked
     hipMemcpyAsync, prepare overlapped on the compute stream): a job
