@@ -168,15 +168,16 @@ def main():
     n_entries = sum(i.nbytes // 16 for runs in job_runs for d, i in runs)
 
     if len(job_runs) > 1:
-        # independent jobs batched into ONE launch set (job-local ranks/
-        # winner flags, outputs sliced per job — dbeel_gpu_job_create_
-        # batched). One kernel pipeline per step replaces 8 per-stream
-        # pipelines' launch storms; set DBEEL_BENCH_STREAMS=1 for the
-        # old one-HIP-stream-per-job mode.
-        if os.environ.get("DBEEL_BENCH_STREAMS") == "1":
-            jobs = [dbeel_amd.Job(runs, device=device) for runs in job_runs]
-        else:
+        # independent jobs, one HIP stream each (one dbeel shard per job —
+        # SURVEY.md §8e); ctypes releases the GIL during the blocking
+        # engine calls. DBEEL_BENCH_BATCHED=1 runs them as ONE batched
+        # launch set instead (dbeel_gpu_job_create_batched) — measured
+        # equivalent on cfg4 (6.83 vs 6.69 ms: the work, not the launch
+        # pattern, bounds this shape), kept for A/B.
+        if os.environ.get("DBEEL_BENCH_BATCHED") == "1":
             jobs = [dbeel_amd.engine.BatchJob(job_runs, device=device)]
+        else:
+            jobs = [dbeel_amd.Job(runs, device=device) for runs in job_runs]
     else:
         jobs = [dbeel_amd.Job(job_runs[0], device=device)]
 

@@ -449,6 +449,18 @@ class Job:
             self._h = None
             self._keepalive = None
 
+    def __enter__(self):
+        return self
+
+    def __exit__(self, *a):
+        self.close()
+
+    def __del__(self):
+        try:
+            self.close()
+        except Exception:
+            pass
+
 
 class BatchJob(Job):
     """Batched INDEPENDENT jobs in one launch set (one dbeel shard each,
@@ -498,14 +510,3 @@ class BatchJob(Job):
             self._lib.dbeel_gpu_result_free(ctypes.byref(res))
         return data, index, n
 
-    def __enter__(self):
-        return self
-
-    def __exit__(self, *a):
-        self.close()
-
-    def __del__(self):
-        try:
-            self.close()
-        except Exception:
-            pass
