@@ -28,13 +28,16 @@ def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--rounds", type=int, default=12)
     ap.add_argument("--scale", type=float, default=1.0)
-    ap.add_argument("--workload", default="cfg3", choices=["cfg3", "cfg5"])
+    ap.add_argument("--workload", default="cfg3",
+                    choices=["cfg2", "cfg3", "cfg5"])
     ap.add_argument("--grids", default="",
                     help="comma list of DBEEL_COPY_GRID caps to sweep "
                          "(per variant); empty = default 4096")
     args = ap.parse_args()
 
-    if args.workload == "cfg5":
+    if args.workload == "cfg2":
+        runs = make_config("cfg2", scale=args.scale)
+    elif args.workload == "cfg5":
         from dbeel_amd.genruns import CONFIGS, make_runs_varkey
 
         cfg = dict(CONFIGS["cfg5"])
