@@ -79,12 +79,40 @@ def random_runs(rng):
     return runs
 
 
+def random_scan_filter(rng):
+    kw = {}
+    if rng.random() < 0.5:
+        kw["start_key"] = bytes(rng.integers(0, 256,
+                                             int(rng.integers(0, 8)),
+                                             dtype=np.uint8))
+    if rng.random() < 0.5:
+        kw["end_key"] = bytes(rng.integers(0, 256,
+                                           int(rng.integers(0, 8)),
+                                           dtype=np.uint8))
+    if rng.random() < 0.7:
+        n = int(rng.integers(1, 4))
+        kw["hash_ranges"] = [
+            (int(rng.integers(0, 2**32)), int(rng.integers(0, 2**32)))
+            for _ in range(n)
+        ]
+    return kw
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--trials", type=int, default=100)
     ap.add_argument("--seed", type=int, default=1)
     ap.add_argument("--device", type=int, default=0)
+    ap.add_argument("--scan", action="store_true",
+                    help="also fuzz dbeel_gpu_scan vs the AsyncIter model")
     args = ap.parse_args()
+
+    if args.scan:
+        sys.path.insert(0, os.path.join(
+            os.path.dirname(os.path.abspath(__file__)), "..", "tests"))
+        from pymm3 import scan_model
+
+        from dbeel_amd.engine import scan as gpu_scan
 
     failures = 0
     for t in range(args.trials):
@@ -98,6 +126,13 @@ def main():
                 failures += 1
                 print(f"TRIAL {t} keep={keep} MISMATCH: "
                       f"n {gn} vs {on}, index {gi == oi}, data {gd == od}")
+        if args.scan:
+            kw = random_scan_filter(rng)
+            sd, si, sn = gpu_scan(runs, device=args.device, **kw)
+            md, mi, mn = scan_model(runs, **kw)
+            if (sd, si, sn) != (md, mi, mn):
+                failures += 1
+                print(f"TRIAL {t} SCAN MISMATCH {kw}: n {sn} vs {mn}")
         if (t + 1) % 25 == 0:
             print(f"{t + 1}/{args.trials} trials OK")
     if failures:
