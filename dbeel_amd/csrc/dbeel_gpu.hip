@@ -1764,12 +1764,14 @@ static void launch_copy(hipStream_t s, const uint8_t* out_index,
     if (variant < 0) {
         uint64_t avg = n_surv ? total_out / n_surv : 0;
         if (avg > 2048) {
-            variant = 0;
+            variant = 0; /* 4 KiB-class: more blocks, small windows */
             if (!gcap) gcap = 16384;
         } else if (avg >= 512) {
-            variant = 4;
+            variant = 4; /* ~1 KiB-class: 64 KiB windows */
         } else {
-            variant = 0;
+            variant = 3; /* sub-512-B: 512thr x 32 KiB + more blocks
+                            (cfg2 A/B: 0.516 -> 0.467 ms) */
+            if (!gcap) gcap = 16384;
         }
     }
     if (!gcap) gcap = 4096;
