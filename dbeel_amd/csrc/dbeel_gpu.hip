@@ -1480,14 +1480,22 @@ extern "C" int dbeel_gpu_job_fetch_job(dbeel_gpu_job* job, size_t job_idx,
         set_err("host malloc failed");
         return DBEEL_ERR_OOM;
     }
+    hipError_t ce = hipSuccess;
     if (dlen)
-        HIP_CHECK(hipMemcpyAsync(out->data, job->d_outdata + off0, dlen,
-                                 hipMemcpyDeviceToHost, s));
-    if (nsurv)
-        HIP_CHECK(hipMemcpyAsync(out->index,
-                                 job->d_outindex + (uint64_t)pos0 * 16,
-                                 nsurv * 16, hipMemcpyDeviceToHost, s));
-    HIP_CHECK(hipStreamSynchronize(s));
+        ce = hipMemcpyAsync(out->data, job->d_outdata + off0, dlen,
+                            hipMemcpyDeviceToHost, s);
+    if (ce == hipSuccess && nsurv)
+        ce = hipMemcpyAsync(out->index,
+                            job->d_outindex + (uint64_t)pos0 * 16,
+                            nsurv * 16, hipMemcpyDeviceToHost, s);
+    if (ce == hipSuccess) ce = hipStreamSynchronize(s);
+    if (ce != hipSuccess) {
+        free(out->data);
+        free(out->index);
+        memset(out, 0, sizeof *out);
+        set_err("fetch_job D2H failed: %s", hipGetErrorString(ce));
+        return DBEEL_ERR_HIP;
+    }
     /* rebase offsets to the job's own file */
     for (uint64_t i = 0; i < nsurv; i++) {
         uint64_t o = ld_u64_host(out->index + i * 16) - off0;

@@ -484,6 +484,22 @@ def test_sliced_compaction_parity(engine):
         assert gd == od
 
 
+def test_sliced_compaction_varkey_parity(engine):
+    """Sliced compaction with VARIABLE-length msgpack keys: host pivot
+    selection and per-run binary search must respect raw-byte key order
+    for ragged keys too (cfg5-style runs, forced multi-slice)."""
+    from dbeel_amd.engine import compact_sliced
+    from dbeel_amd.genruns import make_runs_varkey
+
+    runs = make_runs_varkey(6, 5_000, value_size=512, overlap_frac=0.4,
+                            tombstone_frac=0.1, seed=0xBEEF)
+    total = sum(d.nbytes + i.nbytes for d, i in runs)
+    od, oi, on = oracle.compact(runs, keep_tombstones=False)
+    gd, gi, gn = compact_sliced(runs, keep_tombstones=False, device=0,
+                                max_resident_bytes=total // 5)
+    assert (gn, gi, gd) == (on, oi, od)
+
+
 def test_resident_job_repeatable(engine):
     """Job API: repeated runs on resident inputs give identical results and
     both keep_tombstones settings work on one upload."""
