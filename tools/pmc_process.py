@@ -129,10 +129,59 @@ def overlap(copy_csv, kernel_csv, out_json):
     print(f"{out_json}: overlap {doc['overlap_fraction']}")
 
 
+def stalls(counter_csv, out_json, workload):
+    """Wave-state breakdown per kernel from one SQ-counter pass:
+    SQ_WAVE_CYCLES = parked (SQ_WAIT_ANY: waitcnt/barrier) + issue-stall
+    (SQ_WAIT_INST_ANY) + issuing (SQ_ACTIVE_INST_ANY), disjoint
+    (MI355X_MICROARCH.md SQ table). parked_frac near 1 means waves sit on
+    memory; issue/active-dominated means the kernel is issue-bound."""
+    per = {}
+    with open(counter_csv) as f:
+        for row in csv.DictReader(f):
+            name = kshort(row["Kernel_Name"])
+            if name.startswith("__amd"):
+                continue
+            per.setdefault(name, {}).setdefault(
+                row["Counter_Name"], []).append(float(row["Counter_Value"]))
+    doc = {"_doc": ("Per-kernel wave-state breakdown (one rocprofv3 --pmc "
+                    "pass: SQ_WAVE_CYCLES / SQ_WAIT_ANY / SQ_WAIT_INST_ANY "
+                    "/ SQ_ACTIVE_INST_ANY, quad-cycle units, averaged over "
+                    "launches). parked = waiting on waitcnt/barrier "
+                    "(memory); issue_stall = instruction-issue stalls; "
+                    "active = issuing."),
+           "workload": workload, "kernels": {}}
+    for name, c in sorted(per.items()):
+        def avg(k):
+            v = c.get(k, [])
+            return sum(v) / len(v) if v else 0.0
+        wave = avg("SQ_WAVE_CYCLES")
+        if not wave:
+            continue
+        doc["kernels"][name] = {
+            "wave_cycles": int(wave),
+            "parked_frac": round(avg("SQ_WAIT_ANY") / wave, 4),
+            "issue_stall_frac": round(avg("SQ_WAIT_INST_ANY") / wave, 4),
+            "active_frac": round(avg("SQ_ACTIVE_INST_ANY") / wave, 4),
+        }
+    json.dump(doc, open(out_json, "w"), indent=1)
+    kc = doc["kernels"].get("k_copy", {})
+    print(f"{out_json}: k_copy parked={kc.get('parked_frac')} "
+          f"issue={kc.get('issue_stall_frac')} "
+          f"active={kc.get('active_frac')}")
+
+
 if __name__ == "__main__":
     base = sys.argv[1] if len(sys.argv) > 1 else "gpurun_out"
-    traffic(f"{base}/r02_cfg3_fetch.csv", f"{base}/r02_cfg3_write.csv",
-            "profiles/r02_traffic_cfg3.json")
-    overlap(f"{base}/ing_memory_copy_trace.csv",
-            f"{base}/ing_kernel_trace.csv",
-            "profiles/r02_ingest_overlap.json")
+    import os
+
+    if os.path.exists(f"{base}/r02b_cfg3_fetch.csv"):
+        traffic(f"{base}/r02b_cfg3_fetch.csv", f"{base}/r02b_cfg3_write.csv",
+                "profiles/r02_traffic_cfg3.json")
+    if os.path.exists(f"{base}/ing_memory_copy_trace.csv"):
+        overlap(f"{base}/ing_memory_copy_trace.csv",
+                f"{base}/ing_kernel_trace.csv",
+                "profiles/r02_ingest_overlap.json")
+    for w in ("cfg3", "cfg5"):
+        p = f"{base}/r02_{w}_stalls.csv"
+        if os.path.exists(p):
+            stalls(p, f"profiles/r02_{w}_stalls.json", w)
