@@ -6,32 +6,46 @@
  * the reference's heap loop:
  *
  *   k_prepare    : validates every entry (bounds + bincode field
- *                  cross-check; corrupt input errors loudly where the
- *                  reference silently truncates) and extracts the dense
- *                  key-prefix (8 B) and aux (64 B: klen/ts/first-40-key-
- *                  bytes) arrays the merge runs on.
+ *                  cross-check + EntryWriter density; corrupt input
+ *                  errors loudly where the reference silently truncates)
+ *                  and extracts the dense key-prefix (8 B) and tiered
+ *                  suffix-aux arrays (16/48/64 B: key bytes 8.., klen,
+ *                  staged i128 ts in the duplicate-prone tiers) the
+ *                  merge runs on. Rangeable for streamed ingest.
  *   k_corank     : block-cooperative merge-path co-ranking — for every
- *                  run pair, diagonal-partitioned 4096-position windows;
- *                  blocks stage both prefix segments into LDS coalesced
- *                  and walk CORANK_STEPS merged positions per thread,
- *                  recording
- *                  crossranks (order = key bytes asc, timestamp i128 asc,
- *                  run index asc — lsm_tree.rs:52-71 + mod.rs:75-81) and
+ *                  run pair WITHIN a job, diagonal-partitioned
+ *                  4096-position windows; blocks stage both prefix
+ *                  segments into LDS coalesced and walk CORANK_STEPS
+ *                  merged positions per thread, recording crossranks
+ *                  (order = key bytes asc, timestamp i128 asc, run index
+ *                  asc — lsm_tree.rs:52-71 + mod.rs:75-81) and
  *                  newest-wins supersession flags (lsm_tree.rs:1041-1044).
- *   k_rankreduce : crossranks -> global rank; winner/tombstone rules;
- *                  strict-sortedness check (flush invariant,
- *                  lsm_tree.rs:925-946); one 32-B rank-indexed record.
- *   scans        : rocPRIM exclusive scans (transform iterators) ->
- *                  survivor byte offsets + positions.
+ *   k_rankreduce : crossranks -> job-local rank (+ job entry base);
+ *                  winner/tombstone rules; strict-sortedness check
+ *                  (flush invariant, lsm_tree.rs:925-946); one 16-B
+ *                  rank-indexed record (keep flag in src bit 63).
+ *   scans        : two rocPRIM exclusive scans (transform iterators,
+ *                  plain u64/u32 + plus to stay on the decoupled-
+ *                  lookback fast path) -> survivor byte offsets +
+ *                  positions.
  *   k_emit       : output .index records (offset/key_size/full_size,
  *                  entry_writer.rs:79-87) + compacted source map.
  *   k_winmap/k_copy : verbatim survivor copy, balanced by DESTINATION
- *                  granule (16-B granules per lane, 16-KiB window per
- *                  256-thread block by default, LDS granule->entry map;
- *                  geometry variants for A/B) so throughput
- *                  is independent of entry size; aligned non-temporal
- *                  16-B stores, unaligned 16-B loads.
+ *                  granule (16-B granules per lane, window/grid geometry
+ *                  picked per survivor size from interleaved A/Bs, LDS
+ *                  granule->entry map) so throughput is independent of
+ *                  entry size; full windows gather granule values into
+ *                  registers before storing; aligned non-temporal 16-B
+ *                  stores, unaligned 16-B loads, boundary granules via
+ *                  16-B two-load blends.
+ *   k_scanflag   : the AsyncIter migration scan filter (murmur3 hash
+ *                  ranges + key bounds — lsm_tree.rs:141-282,
+ *                  tasks/migration.rs:54-60).
  *   k_encode_*   : the memtable-flush run encoder (lsm_tree.rs:925-946).
+ *   host side    : resident jobs, batched independent jobs (one launch
+ *                  set, per-job output slicing), streamed pinned ingest
+ *                  with copy/compute overlap, sliced compaction for
+ *                  inputs larger than HBM, device bloom build.
  *
  * All integer/byte work, HBM-bandwidth bound; MFMA unused by design
  * (BASELINE.json north_star).

@@ -484,6 +484,21 @@ def test_sliced_compaction_parity(engine):
         assert gd == od
 
 
+def test_sliced_compaction_midscale_parity(engine):
+    """Sliced compaction at ~0.9 GB (cfg3-shaped, forced 4 slices): the
+    pivot/binary-search machinery and cross-slice offset rebasing stay
+    bit-exact when slices span many copy windows."""
+    from dbeel_amd.engine import compact_sliced
+
+    runs = make_runs(8, 100_000, 32, 1024, overlap_frac=0.5,
+                     tombstone_frac=0.05, seed=0xDBEE1)
+    total = sum(d.nbytes + i.nbytes for d, i in runs)
+    od, oi, on = oracle.compact(runs, keep_tombstones=False)
+    gd, gi, gn = compact_sliced(runs, keep_tombstones=False, device=0,
+                                max_resident_bytes=total // 4)
+    assert (gn, gi, gd) == (on, oi, od)
+
+
 def test_sliced_compaction_varkey_parity(engine):
     """Sliced compaction with VARIABLE-length msgpack keys: host pivot
     selection and per-run binary search must respect raw-byte key order
