@@ -644,9 +644,17 @@ __global__ void k_rankreduce(RunsDesc R, const uint64_t* pfx,
     }
 }
 
-/* transform-iterator functors for the two survivor scans (plain
- * arithmetic types + rocprim::plus keep rocPRIM on its decoupled-
- * lookback fast path — a fused custom-type scan measured 10x slower) */
+/* transform-iterator functors for the survivor scans (plain arithmetic
+ * types + rocprim::plus keep rocPRIM on its decoupled-lookback fast
+ * path — a fused custom-STRUCT scan measured 10x slower). When the job
+ * fits the packing bounds (bytes < 2^38 = 256 GB, entries < 2^26 = 67M
+ * — every BASELINE shape does), ONE u64 scan carries BOTH running sums:
+ * survivor bytes in the low 38 bits, survivor count in the high 26
+ * (sums stay in-field, so lane-wise u64 addition never carries across);
+ * that halves the scan's passes over rrec. Larger jobs fall back to two
+ * scans. */
+#define PK_SHIFT 38
+#define PK_MASK ((1ull << PK_SHIFT) - 1)
 struct RankRecSize {
     __device__ uint64_t operator()(const RankRec& r) const {
         return (r.src & RR_KEEP) ? (uint64_t)r.full_size : 0ull;
@@ -655,6 +663,13 @@ struct RankRecSize {
 struct RankRecFlag {
     __device__ uint32_t operator()(const RankRec& r) const {
         return (r.src & RR_KEEP) ? 1u : 0u;
+    }
+};
+struct RankRecPacked {
+    __device__ uint64_t operator()(const RankRec& r) const {
+        return (r.src & RR_KEEP)
+                   ? ((uint64_t)r.full_size | (1ull << PK_SHIFT))
+                   : 0ull;
     }
 };
 
@@ -673,6 +688,28 @@ __global__ void k_emit(RunsDesc R, const RankRec* rrec,
         uint64_t p = pos[g];
         uint8_t* rec = out_index + p * 16;
         uint64_t off = dst_off[g];
+        __builtin_memcpy(rec, &off, 8);
+        __builtin_memcpy(rec + 8, &m.key_size, 4);
+        __builtin_memcpy(rec + 12, &m.full_size, 4);
+        src_map[p] = (uint64_t)(R.data[(m.src >> 48) & 0x7FFF] +
+                                (m.src & RR_OFF_MASK));
+    }
+}
+
+/* k_emit for the packed single-scan mode: offsets and positions decode
+ * from one u64 (low 38 bits = byte offset, high 26 = survivor index). */
+__global__ void k_emit_packed(RunsDesc R, const RankRec* rrec,
+                              const uint64_t* packed, uint64_t total,
+                              uint8_t* out_index, uint64_t* src_map) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t g = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         g < total; g += stride) {
+        RankRec m = rrec[g];
+        if (!(m.src & RR_KEEP)) continue;
+        uint64_t v = packed[g];
+        uint64_t p = v >> PK_SHIFT;
+        uint8_t* rec = out_index + p * 16;
+        uint64_t off = v & PK_MASK;
         __builtin_memcpy(rec, &off, 8);
         __builtin_memcpy(rec + 8, &m.key_size, 4);
         __builtin_memcpy(rec + 12, &m.full_size, 4);
@@ -1155,6 +1192,7 @@ struct dbeel_gpu_job {
     uint64_t input_bytes = 0;
     std::vector<uint64_t> job_gbase; /* entry base per job (+ total) */
     /* last-run results */
+    int last_scan_packed = 0; /* d_dstoff holds packed (pos<<38|off) */
     uint64_t out_data_len = 0;
     uint64_t out_entries = 0;
     bool have_result = false;
@@ -1466,23 +1504,37 @@ extern "C" int dbeel_gpu_job_fetch_job(dbeel_gpu_job* job, size_t job_idx,
     memset(out, 0, sizeof *out);
     uint64_t g0 = job->job_gbase[job_idx];
     uint64_t g1 = job->job_gbase[job_idx + 1];
-    /* boundary offsets/positions: tiny D2H reads */
+    /* boundary offsets/positions: tiny D2H reads (packed scan mode
+     * carries both in one u64 — decode below) */
     uint64_t off0 = 0, off1 = job->out_data_len;
     uint32_t pos0 = 0, pos1 = (uint32_t)job->out_entries;
     hipStream_t s = job->stream;
+    int pk = job->last_scan_packed;
     if (g0 > 0) {
         HIP_CHECK(hipMemcpyAsync(&off0, job->d_dstoff + g0, 8,
                                  hipMemcpyDeviceToHost, s));
-        HIP_CHECK(hipMemcpyAsync(&pos0, job->d_pos + g0, 4,
-                                 hipMemcpyDeviceToHost, s));
+        if (!pk)
+            HIP_CHECK(hipMemcpyAsync(&pos0, job->d_pos + g0, 4,
+                                     hipMemcpyDeviceToHost, s));
     }
     if (g1 < job->total_entries) {
         HIP_CHECK(hipMemcpyAsync(&off1, job->d_dstoff + g1, 8,
                                  hipMemcpyDeviceToHost, s));
-        HIP_CHECK(hipMemcpyAsync(&pos1, job->d_pos + g1, 4,
-                                 hipMemcpyDeviceToHost, s));
+        if (!pk)
+            HIP_CHECK(hipMemcpyAsync(&pos1, job->d_pos + g1, 4,
+                                     hipMemcpyDeviceToHost, s));
     }
     HIP_CHECK(hipStreamSynchronize(s));
+    if (pk) {
+        if (g0 > 0) {
+            pos0 = (uint32_t)(off0 >> PK_SHIFT);
+            off0 &= PK_MASK;
+        }
+        if (g1 < job->total_entries) {
+            pos1 = (uint32_t)(off1 >> PK_SHIFT);
+            off1 &= PK_MASK;
+        }
+    }
     uint64_t dlen = off1 - off0;
     uint64_t nsurv = (uint64_t)pos1 - pos0;
     out->data = (uint8_t*)malloc(dlen ? dlen : 1);
@@ -1866,24 +1918,47 @@ extern "C" int dbeel_gpu_job_run(dbeel_gpu_job* job, int keep_tombstones,
         HIP_CHECK(hipEventRecord(job->ev[6], s));
     }
     HIP_CHECK(hipEventRecord(job->ev[1], s));
+    bool packed = n < (1ull << 26) &&
+                  job->total_data_bytes < (1ull << PK_SHIFT);
+    if (const char* m = getenv("DBEEL_SCAN_MODE")) {
+        if (!strcmp(m, "two")) packed = false;
+    }
     if (n) {
         size_t tmp = job->scantmp_bytes;
-        (void)rocprim::exclusive_scan(
-            job->d_scantmp, tmp,
-            rocprim::make_transform_iterator(job->d_rank, RankRecSize{}),
-            job->d_dstoff, (uint64_t)0, n, rocprim::plus<uint64_t>(), s);
-        tmp = job->scantmp_bytes;
-        (void)rocprim::exclusive_scan(
-            job->d_scantmp, tmp,
-            rocprim::make_transform_iterator(job->d_rank, RankRecFlag{}),
-            job->d_pos, (uint32_t)0, n, rocprim::plus<uint32_t>(), s);
+        if (packed) {
+            (void)rocprim::exclusive_scan(
+                job->d_scantmp, tmp,
+                rocprim::make_transform_iterator(job->d_rank,
+                                                 RankRecPacked{}),
+                job->d_dstoff, (uint64_t)0, n, rocprim::plus<uint64_t>(),
+                s);
+        } else {
+            (void)rocprim::exclusive_scan(
+                job->d_scantmp, tmp,
+                rocprim::make_transform_iterator(job->d_rank,
+                                                 RankRecSize{}),
+                job->d_dstoff, (uint64_t)0, n, rocprim::plus<uint64_t>(),
+                s);
+            tmp = job->scantmp_bytes;
+            (void)rocprim::exclusive_scan(
+                job->d_scantmp, tmp,
+                rocprim::make_transform_iterator(job->d_rank,
+                                                 RankRecFlag{}),
+                job->d_pos, (uint32_t)0, n, rocprim::plus<uint32_t>(), s);
+        }
     }
     HIP_CHECK(hipEventRecord(job->ev[2], s));
     if (n) {
         uint32_t grid = pick_grid(n, 256);
-        hipLaunchKernelGGL(k_emit, dim3(grid), dim3(256), 0, s, job->desc,
-                           job->d_rank, job->d_dstoff, job->d_pos, n,
-                           job->d_outindex, job->d_srcmap);
+        if (packed)
+            hipLaunchKernelGGL(k_emit_packed, dim3(grid), dim3(256), 0, s,
+                               job->desc, job->d_rank, job->d_dstoff, n,
+                               job->d_outindex, job->d_srcmap);
+        else
+            hipLaunchKernelGGL(k_emit, dim3(grid), dim3(256), 0, s,
+                               job->desc, job->d_rank, job->d_dstoff,
+                               job->d_pos, n, job->d_outindex,
+                               job->d_srcmap);
     }
     HIP_CHECK(hipEventRecord(job->ev[3], s));
 
@@ -1897,8 +1972,9 @@ extern "C" int dbeel_gpu_job_run(dbeel_gpu_job* job, int keep_tombstones,
                                  hipMemcpyDeviceToHost, s));
         HIP_CHECK(hipMemcpyAsync(&last_off, job->d_dstoff + (n - 1), 8,
                                  hipMemcpyDeviceToHost, s));
-        HIP_CHECK(hipMemcpyAsync(&last_pos, job->d_pos + (n - 1), 4,
-                                 hipMemcpyDeviceToHost, s));
+        if (!packed)
+            HIP_CHECK(hipMemcpyAsync(&last_pos, job->d_pos + (n - 1), 4,
+                                     hipMemcpyDeviceToHost, s));
     }
     HIP_CHECK(hipMemcpyAsync(&err, job->d_err, 4, hipMemcpyDeviceToHost, s));
     HIP_CHECK(hipStreamSynchronize(s));
@@ -1909,6 +1985,11 @@ extern "C" int dbeel_gpu_job_run(dbeel_gpu_job* job, int keep_tombstones,
                     : "corrupt entry/index record");
         return DBEEL_ERR_CORRUPT;
     }
+    if (packed) {
+        last_pos = (uint32_t)(last_off >> PK_SHIFT);
+        last_off &= PK_MASK;
+    }
+    job->last_scan_packed = packed ? 1 : 0;
     uint64_t last_kept = (n && (last_rec.src & RR_KEEP)) ? 1 : 0;
     uint64_t total_out = last_off + (last_kept ? last_rec.full_size : 0);
     uint64_t n_surv = (uint64_t)last_pos + last_kept;
@@ -2389,6 +2470,7 @@ extern "C" int dbeel_gpu_scan(const dbeel_run_view* runs, size_t n_runs,
     if (e == hipSuccess) e = hipStreamSynchronize(s);
     if (e == hipSuccess) e = hipGetLastError();
 
+    job->last_scan_packed = 0;
     int ret = DBEEL_OK;
     if (e != hipSuccess) {
         set_err("scan failed: %s", hipGetErrorString(e));

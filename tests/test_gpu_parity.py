@@ -339,6 +339,38 @@ def test_migration_scan_at_scale(engine):
     assert (gn, gi, gd) == (mn, mi, md)
 
 
+def test_scan_modes_agree(engine):
+    """The packed single survivor scan (bytes|count in one u64) and the
+    two-scan fallback must produce identical results, including batched
+    per-job slicing whose boundary reads decode the packed values."""
+    import os
+
+    from dbeel_amd.engine import BatchJob
+
+    jobs = [
+        make_runs(4, 8_000, 16, 256, overlap_frac=0.4, tombstone_frac=0.1,
+                  seed=301),
+        make_runs(3, 6_000, 32, 512, overlap_frac=0.6, tombstone_frac=0.2,
+                  seed=302),
+    ]
+    results = {}
+    for mode in ("packed", "two"):
+        os.environ["DBEEL_SCAN_MODE"] = mode
+        try:
+            with BatchJob(jobs, device=0) as bj:
+                tot = bj.run(False)[:2]
+                results[mode] = (tot, [bj.fetch_job(j)
+                                       for j in range(len(jobs))])
+        finally:
+            os.environ.pop("DBEEL_SCAN_MODE", None)
+    assert results["packed"] == results["two"]
+    # and both match the oracle
+    for j, runs in enumerate(jobs):
+        od, oi, on = oracle.compact(runs, keep_tombstones=False)
+        gd, gi, gn = results["packed"][1][j]
+        assert (gn, gi, gd) == (on, oi, od)
+
+
 def test_batched_jobs_parity(engine):
     """Batched independent jobs (one launch set, BASELINE configs[3]'s
     8-jobs-per-GPU shape): each job's sliced-out result must be
