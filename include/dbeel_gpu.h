@@ -160,6 +160,22 @@ typedef struct dbeel_gpu_job dbeel_gpu_job;
 
 int dbeel_gpu_job_create(const dbeel_run_view* runs, size_t n_runs,
                          int device, dbeel_gpu_job** out_job);
+/* Batched INDEPENDENT jobs in one launch set (one dbeel shard each —
+ * BASELINE configs[3]'s many-jobs-per-GPU shape): runs[] holds every
+ * job's runs back to back, runs_per_job[] their counts (each >= 1,
+ * summing to n_runs). Jobs never interact: ranks, crossrank slots and
+ * newest-wins flags are job-local, and a key present in two jobs
+ * survives in both. Outputs land concatenated in job order;
+ * dbeel_gpu_job_fetch_job slices job `job_idx`'s result back out with
+ * index offsets rebased to its own run file. job_run/job_ingest/
+ * job_fetch work on batched jobs unchanged (fetch returns the
+ * concatenation). */
+int dbeel_gpu_job_create_batched(const dbeel_run_view* runs, size_t n_runs,
+                                 const uint32_t* runs_per_job,
+                                 size_t n_jobs, int device,
+                                 dbeel_gpu_job** out_job);
+int dbeel_gpu_job_fetch_job(dbeel_gpu_job* job, size_t job_idx,
+                            dbeel_compact_result* out);
 /* Runs the pipeline; fills t (may be NULL) and the output sizes. */
 int dbeel_gpu_job_run(dbeel_gpu_job* job, int keep_tombstones,
                       uint64_t* out_data_len, uint64_t* out_entries,

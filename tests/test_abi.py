@@ -18,10 +18,13 @@ HEADER_SYMBOLS = [
     "dbeel_gpu_last_error",
     "dbeel_gpu_encode_run",
     "dbeel_gpu_job_create",
+    "dbeel_gpu_job_create_batched",
     "dbeel_gpu_job_run",
     "dbeel_gpu_job_fetch",
+    "dbeel_gpu_job_fetch_job",
     "dbeel_gpu_job_ingest",
     "dbeel_gpu_job_destroy",
+    "dbeel_gpu_scan",
     "dbeel_gpu_pin_host",
     "dbeel_gpu_unpin_host",
 ]
