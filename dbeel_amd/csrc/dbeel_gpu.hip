@@ -1717,7 +1717,29 @@ extern "C" int dbeel_gpu_job_ingest(dbeel_gpu_job* job,
     for (size_t r = 0; r < n_runs; r++) {
         uint64_t n = job->desc.count[r];
         uint64_t dlen = runs[r].data_len;
-        if (!dlen) continue;
+        if (!dlen) {
+            /* entries with no data bytes are corrupt; the ranged prepare
+             * must still run over them so the corruption is FLAGGED (it
+             * is normally launched per data chunk) */
+            if (n) {
+                uint64_t g0 = job->desc.entry_base[r];
+                uint64_t g1 = g0 + n;
+                uint32_t grid = pick_grid(n, 256);
+                switch (job->aux_kind) {
+#define PREP_EMPTY(KB, TS)                                                  \
+    case (KB == 12 ? 0 : (KB == 24 ? 1 : 2)):                               \
+        hipLaunchKernelGGL((k_prepare<KB, TS>), dim3(grid), dim3(256), 0,   \
+                           ks, job->desc, g0, g1, job->d_pfx,               \
+                           (AuxT<KB, TS>*)job->d_aux, job->d_err);          \
+        break;
+                    PREP_EMPTY(12, false)
+                    PREP_EMPTY(24, true)
+                    PREP_EMPTY(40, true)
+#undef PREP_EMPTY
+                }
+            }
+            continue;
+        }
         const uint8_t* hidx = runs[r].index;
         uint64_t e0 = 0, b0 = 0;
         while (b0 < dlen) {
@@ -1743,6 +1765,14 @@ extern "C" int dbeel_gpu_job_ingest(dbeel_gpu_job* job,
                 if (b1 <= b0) { /* one giant entry spans the chunk */
                     e1 = e0 + 1;
                     b1 = (e1 >= n) ? dlen : ld_u64_host(hidx + e1 * 16);
+                }
+                /* corrupt (non-monotone) offsets could still point
+                 * backwards or past the run: clamp to a safe single
+                 * tail chunk — the on-device validation then reports
+                 * CORRUPT instead of this loop wrapping a copy size */
+                if (b1 <= b0 || b1 > dlen) {
+                    e1 = n;
+                    b1 = dlen;
                 }
             }
             ING_CHECK(hipMemcpyAsync(
@@ -2549,13 +2579,21 @@ static int host_key_cmp(const uint8_t* a, uint64_t la, const uint8_t* b,
     return la < lb ? -1 : (la > lb ? 1 : 0);
 }
 
-/* key view of entry i of a host run */
+/* key view of entry i of a host run; NULL when the index record points
+ * outside the run's data (corrupt input must not cause host OOB reads
+ * during pivot selection — the device validation would catch it later,
+ * but the slicer reads keys FIRST) */
 static inline const uint8_t* host_entry_key(const dbeel_run_view* run,
                                             uint64_t i, uint64_t* klen) {
     const uint8_t* rec = run->index + i * 16;
     uint64_t off = ld_u64_host(rec);
-    uint32_t key_size;
+    uint32_t key_size, full_size;
     memcpy(&key_size, rec + 8, 4);
+    memcpy(&full_size, rec + 12, 4);
+    if (key_size < 8 || full_size < 32 ||
+        (uint64_t)key_size + 24 > full_size ||
+        off + full_size > run->data_len)
+        return NULL;
     *klen = key_size - 8;
     return run->data + off + 8;
 }
@@ -2605,6 +2643,10 @@ extern "C" int dbeel_gpu_compact_sliced(const dbeel_run_view* runs,
         uint64_t klen;
         const uint8_t* k =
             host_entry_key(&runs[big_run], j * max_count / S, &klen);
+        if (!k) {
+            set_err("corrupt index record in run %zu", big_run);
+            return DBEEL_ERR_CORRUPT;
+        }
         pivots.push_back({k, klen});
     }
     /* per run: slice boundary entry indices (first entry >= pivot) */
@@ -2618,6 +2660,10 @@ extern "C" int dbeel_gpu_compact_sliced(const dbeel_run_view* runs,
                 uint64_t mid = (lo + hi) >> 1;
                 uint64_t kl;
                 const uint8_t* k = host_entry_key(&runs[r], mid, &kl);
+                if (!k) {
+                    set_err("corrupt index record in run %zu", r);
+                    return DBEEL_ERR_CORRUPT;
+                }
                 if (host_key_cmp(k, kl, pv.first, pv.second) < 0)
                     lo = mid + 1;
                 else
