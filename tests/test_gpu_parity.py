@@ -516,6 +516,58 @@ def test_sliced_compaction_parity(engine):
         assert gd == od
 
 
+def test_ingest_rejects_corrupt_offsets(engine):
+    """Streamed ingest with NON-MONOTONE index offsets (corrupt input)
+    must report CORRUPT, never wrap a chunk copy or accept the data."""
+    import os
+
+    from dbeel_amd.engine import DbeelGpuError
+
+    runs = make_runs(2, 2_000, 16, 64, seed=11)
+    bad = [(d.copy(), i.copy()) for d, i in runs]
+    # swap two offset fields in run 0's index -> offsets not monotone
+    idx = bad[0][1]
+    rec0 = idx[0:8].copy()
+    idx[0:8] = idx[160:168]
+    idx[160:168] = rec0
+    os.environ["DBEEL_STREAM_CHUNK_MB"] = "1"
+    try:
+        with engine.Job(runs, device=0) as job:
+            with pytest.raises(DbeelGpuError) as ei:
+                job.ingest(bad)
+            assert ei.value.code == 2  # CORRUPT
+            # the job stays usable: re-ingest good data, run, verify
+            job.ingest(runs)
+            job.run(False)
+            gd, gi, gn = job.fetch()
+            od, oi, on = oracle.compact(runs, keep_tombstones=False)
+            assert (gn, gi, gd) == (on, oi, od)
+    finally:
+        os.environ.pop("DBEEL_STREAM_CHUNK_MB", None)
+
+
+def test_sliced_rejects_corrupt_index(engine):
+    """Sliced compaction reads host key bytes during pivot selection —
+    a record pointing past the run's data must yield CORRUPT, not an
+    out-of-bounds host read."""
+    import struct
+
+    from dbeel_amd.engine import DbeelGpuError, compact_sliced
+
+    runs = [list(r) for r in make_runs(3, 4_000, 16, 64, seed=12)]
+    bad_idx = runs[0][1].copy()
+    # entry 2000 is the FIRST binary-search probe (mid of 0..4000) and a
+    # candidate pivot — point its offset far beyond data_len
+    bad_idx[2000 * 16 : 2000 * 16 + 8] = np.frombuffer(
+        struct.pack("<Q", 1 << 40), dtype=np.uint8)
+    runs[0][1] = bad_idx
+    total = sum(d.nbytes + i.nbytes for d, i in runs)
+    with pytest.raises(DbeelGpuError) as ei:
+        compact_sliced([tuple(r) for r in runs], keep_tombstones=False,
+                       device=0, max_resident_bytes=total // 4)
+    assert ei.value.code == 2  # CORRUPT
+
+
 def test_sliced_compaction_midscale_parity(engine):
     """Sliced compaction at ~0.9 GB (cfg3-shaped, forced 4 slices): the
     pivot/binary-search machinery and cross-slice offset rebasing stay
