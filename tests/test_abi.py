@@ -134,6 +134,44 @@ def test_no_gpu_errors_loudly():
         pass
 
 
+def _build_abi_demo(tmp_path):
+    import subprocess
+
+    out = tmp_path / "abi_demo"
+    subprocess.run(
+        ["gcc", "-O2", "-I", os.path.join(REPO_ROOT, "include"),
+         os.path.join(REPO_ROOT, "tools", "abi_demo.c"),
+         "-L", os.path.join(REPO_ROOT, "dbeel_amd"), "-ldbeel_gpu",
+         "-Wl,-rpath," + os.path.join(REPO_ROOT, "dbeel_amd"),
+         "-o", str(out)],
+        check=True,
+    )
+    return out
+
+
+def test_abi_demo_compiles_with_plain_gcc(tmp_path):
+    """The drop-in boundary is host-language-free: a plain-C consumer
+    (tools/abi_demo.c) compiles with gcc against include/dbeel_gpu.h and
+    links libdbeel_gpu.so — the same shape as dbeel's Rust FFI stub.
+    (Execution needs a GPU; test_abi_demo_runs in the gpu suite.)"""
+    _built()
+    _build_abi_demo(tmp_path)
+
+
+@pytest.mark.gpu
+def test_abi_demo_runs(tmp_path):
+    """The plain-C ABI consumer executes end to end on the GPU: compact
+    (newest-wins + tombstone drop), newest-index-first lookup, full
+    iteration scan."""
+    import subprocess
+
+    _built()
+    out = _build_abi_demo(tmp_path)
+    r = subprocess.run([str(out)], capture_output=True, text=True)
+    assert r.returncode == 0, r.stderr + r.stdout
+    assert "abi_demo OK" in r.stdout
+
+
 def test_oracle_lib_exports():
     import oracle
 
