@@ -43,7 +43,7 @@ def random_runs(rng):
         pool = {bytes(rng.integers(0, 3, int(rng.integers(1, 12)),
                                    dtype=np.uint8))
                 for _ in range(pool_size)}
-    else:  # long keys past the 40-byte aux window
+    else:  # long keys past every staged-suffix aux tier
         pool = {bytes(rng.integers(0, 4, int(rng.integers(41, 129)),
                                    dtype=np.uint8))
                 for _ in range(pool_size)}
