@@ -438,21 +438,21 @@ struct PairDesc {
     uint64_t chunk_base; /* exclusive prefix sum of per-pair windows */
 };
 
+/* Window-corner diagonal searches, one THREAD per chunk — fully
+ * parallel. Previously each corank block ran its start/end corner
+ * searches on 2 threads while the other 254 waited at the barrier
+ * (corank measured 83% wave-parked); precomputing corners removes the
+ * serial section. d_corners[t] = the merged-path `ia` at chunk t's
+ * starting diagonal; a chunk's END corner is the next chunk's start
+ * (same pair) or the pair's (Na, Nb) terminus. */
 template <int KB, bool TS>
-__global__ __launch_bounds__(CORANK_BLOCK) void k_corank(
-    RunsDesc R, const uint64_t* pfx, const AuxT<KB, TS>* aux,
-    const PairDesc* pairs, uint32_t n_pairs, uint64_t total_chunks,
-    uint32_t* cr) {
-    __shared__ uint64_t s_pfx[CORANK_BLOCK_POS + 2];
-    __shared__ uint32_t s_cr[CORANK_BLOCK_POS]; /* staged crossranks:
-        [0..lenA) for run a, [lenA..lenA+lenB) for run b — each block's
-        output ranges are contiguous, so results are staged here and
-        written back coalesced (direct interleaved 4-B stores measured
-        3x write amplification from line RMW thrash) */
-    __shared__ uint64_t s_bounds[4]; /* iaS, ibS, iaE, ibE */
-
-    for (uint64_t t = blockIdx.x; t < total_chunks; t += gridDim.x) {
-        /* largest pair with chunk_base <= t (uniform across the block) */
+__global__ void k_corners(RunsDesc R, const uint64_t* pfx,
+                          const AuxT<KB, TS>* aux, const PairDesc* pairs,
+                          uint32_t n_pairs, uint64_t total_chunks,
+                          uint64_t* corners) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t t = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         t < total_chunks; t += stride) {
         uint32_t plo = 0, phi = n_pairs;
         while (plo < phi) {
             uint32_t mid = (plo + phi) >> 1;
@@ -466,33 +466,63 @@ __global__ __launch_bounds__(CORANK_BLOCK) void k_corank(
         const uint64_t Na = R.count[a], Nb = R.count[b];
         const uint64_t* gpa = pfx + R.entry_base[a];
         const uint64_t* gpb = pfx + R.entry_base[b];
+        uint64_t diag = (t - P.chunk_base) * CORANK_BLOCK_POS;
+        uint64_t slo = diag > Nb ? diag - Nb : 0;
+        uint64_t shi = diag < Na ? diag : Na;
+        while (slo < shi) {
+            uint64_t mid = (slo + shi) >> 1;
+            uint64_t pa = gpa[mid], pb = gpb[diag - mid - 1];
+            int c = (pa != pb)
+                        ? (pa < pb ? -1 : 1)
+                        : cmp_sfx_full<KB, TS>(R, aux, a, mid, b,
+                                               diag - mid - 1);
+            if (c < 0)
+                slo = mid + 1;
+            else
+                shi = mid;
+        }
+        corners[t] = slo;
+    }
+}
+
+template <int KB, bool TS>
+__global__ __launch_bounds__(CORANK_BLOCK) void k_corank(
+    RunsDesc R, const uint64_t* pfx, const AuxT<KB, TS>* aux,
+    const PairDesc* pairs, uint32_t n_pairs, uint64_t total_chunks,
+    const uint64_t* corners, uint32_t* cr) {
+    __shared__ uint64_t s_pfx[CORANK_BLOCK_POS + 2];
+    __shared__ uint32_t s_cr[CORANK_BLOCK_POS]; /* staged crossranks:
+        [0..lenA) for run a, [lenA..lenA+lenB) for run b — each block's
+        output ranges are contiguous, so results are staged here and
+        written back coalesced (direct interleaved 4-B stores measured
+        3x write amplification from line RMW thrash) */
+    for (uint64_t t = blockIdx.x; t < total_chunks; t += gridDim.x) {
+        /* largest pair with chunk_base <= t (uniform across the block) */
+        uint32_t plo = 0, phi = n_pairs;
+        while (plo < phi) {
+            uint32_t mid = (plo + phi) >> 1;
+            if (pairs[mid].chunk_base <= t)
+                plo = mid + 1;
+            else
+                phi = mid;
+        }
+        /* next pair's base bounds this pair's chunk range */
+        uint64_t pair_end_chunk =
+            (plo < n_pairs) ? pairs[plo].chunk_base : total_chunks;
+        const PairDesc P = pairs[plo - 1];
+        const int a = (int)P.a, b = (int)P.b;
+        const uint64_t Na = R.count[a], Nb = R.count[b];
+        const uint64_t* gpa = pfx + R.entry_base[a];
+        const uint64_t* gpb = pfx + R.entry_base[b];
         uint64_t diag0 = (t - P.chunk_base) * CORANK_BLOCK_POS;
         uint64_t diag1 = diag0 + CORANK_BLOCK_POS;
         if (diag1 > Na + Nb) diag1 = Na + Nb;
 
-        /* window corners by global diagonal search (threads 0 and 1) */
-        if (threadIdx.x < 2) {
-            uint64_t diag = threadIdx.x ? diag1 : diag0;
-            uint64_t slo = diag > Nb ? diag - Nb : 0;
-            uint64_t shi = diag < Na ? diag : Na;
-            while (slo < shi) {
-                uint64_t mid = (slo + shi) >> 1;
-                uint64_t pa = gpa[mid], pb = gpb[diag - mid - 1];
-                int c = (pa != pb)
-                            ? (pa < pb ? -1 : 1)
-                            : cmp_sfx_full<KB, TS>(R, aux, a, mid, b,
-                                                   diag - mid - 1);
-                if (c < 0)
-                    slo = mid + 1;
-                else
-                    shi = mid;
-            }
-            s_bounds[threadIdx.x * 2] = slo;            /* ia */
-            s_bounds[threadIdx.x * 2 + 1] = diag - slo; /* ib */
-        }
-        __syncthreads();
-        const uint64_t iaS = s_bounds[0], ibS = s_bounds[1];
-        const uint64_t iaE = s_bounds[2], ibE = s_bounds[3];
+        /* window corners precomputed by k_corners */
+        const uint64_t iaS = corners[t];
+        const uint64_t ibS = diag0 - iaS;
+        const uint64_t iaE = (t + 1 < pair_end_chunk) ? corners[t + 1] : Na;
+        const uint64_t ibE = diag1 - iaE;
         const uint32_t lenA = (uint32_t)(iaE - iaS);
         const uint32_t lenB = (uint32_t)(ibE - ibS);
 
@@ -1199,6 +1229,7 @@ struct dbeel_gpu_job {
     uint32_t n_pairs = 0;
     uint64_t total_chunks = 0;
     uint32_t* d_winp0 = nullptr; /* copy window -> first survivor       */
+    uint64_t* d_corners = nullptr; /* corank chunk start corners        */
     uint64_t total_entries = 0;
     uint64_t total_data_bytes = 0;
     uint64_t input_bytes = 0;
@@ -1420,6 +1451,17 @@ static int job_create_impl(const dbeel_run_view* runs, size_t n_runs,
             }
         job->n_pairs = np;
         job->total_chunks = cbase;
+        {
+            hipError_t _e = hipMalloc(&job->d_corners,
+                                      (cbase ? cbase : 1) * 8);
+            if (_e != hipSuccess) {
+                set_err("corner table alloc failed: %s",
+                        hipGetErrorString(_e));
+                free(hp);
+                dbeel_gpu_job_destroy(job);
+                return DBEEL_ERR_OOM;
+            }
+        }
         if (np) {
             hipError_t _e = hipMalloc(&job->d_pairs, np * sizeof(PairDesc));
             if (_e == hipSuccess)
@@ -1602,6 +1644,7 @@ extern "C" void dbeel_gpu_job_destroy(dbeel_gpu_job* job) {
     hipFree(job->d_cr);
     hipFree(job->d_pairs);
     hipFree(job->d_winp0);
+    hipFree(job->d_corners);
     for (int i = 0; i < 8; i++)
         if (job->ev[i]) hipEventDestroy(job->ev[i]);
     if (job->stream) hipStreamDestroy(job->stream);
@@ -1940,12 +1983,20 @@ extern "C" int dbeel_gpu_job_run(dbeel_gpu_job* job, int keep_tombstones,
                                0, s, job->desc, (uint64_t)0, n, job->d_pfx, \
                                (AuxT<KB, TS>*)job->d_aux, job->d_err);      \
         hipEventRecord(job->ev[6], s);                                      \
-        if (job->n_pairs)                                                   \
+        if (job->n_pairs) {                                                 \
+            hipLaunchKernelGGL((k_corners<KB, TS>),                         \
+                               dim3(pick_grid(job->total_chunks, 256)),     \
+                               dim3(256), 0, s, job->desc, job->d_pfx,      \
+                               (const AuxT<KB, TS>*)job->d_aux,             \
+                               (const PairDesc*)job->d_pairs, job->n_pairs, \
+                               job->total_chunks, job->d_corners);          \
             hipLaunchKernelGGL((k_corank<KB, TS>), dim3((uint32_t)cgrid),   \
                                dim3(CORANK_BLOCK), 0, s, job->desc,         \
                                job->d_pfx, (const AuxT<KB, TS>*)job->d_aux, \
                                (const PairDesc*)job->d_pairs, job->n_pairs, \
-                               job->total_chunks, job->d_cr);               \
+                               job->total_chunks, job->d_corners,           \
+                               job->d_cr);                                  \
+        }                                                                   \
         hipLaunchKernelGGL((k_rankreduce<KB, TS>), dim3(grid), dim3(256),   \
                            0, s, job->desc, job->d_pfx,                     \
                            (const AuxT<KB, TS>*)job->d_aux, job->d_cr,      \
