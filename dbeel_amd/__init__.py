@@ -7,6 +7,20 @@ format and trigger logic live beside it; everything else in dbeel is out of
 scope by contract.
 """
 from . import format  # noqa: F401
-from .engine import Job, DbeelGpuError, compact  # noqa: F401
+from .engine import (  # noqa: F401
+    BatchJob,
+    DbeelGpuError,
+    Job,
+    compact,
+    compact_sliced,
+    encode_run,
+    lookup,
+    pin_host,
+    scan,
+    unpin_host,
+)
 
-__all__ = ["compact", "Job", "DbeelGpuError", "format"]
+__all__ = [
+    "compact", "compact_sliced", "scan", "lookup", "encode_run",
+    "Job", "BatchJob", "pin_host", "unpin_host", "DbeelGpuError", "format",
+]
