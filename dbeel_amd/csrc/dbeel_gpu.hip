@@ -881,6 +881,10 @@ __global__ __launch_bounds__(BLK) void k_copy(
                 const uint8_t* src = (const uint8_t*)s_src[j] +
                                      (gpos - s_off[j]);
                 if (gpos + 16 <= e_end) {
+                    /* plain (cached) loads: a nontemporal-load variant
+                     * measured 15-18% SLOWER (copy 2.15 -> 2.52 ms on
+                     * cfg3) — adjacent granules of the same entry share
+                     * lines through L2 and NT hints forfeit that */
                     __builtin_memcpy(&vv[q], src, 16);
                 } else {
                     /* one entry boundary inside the granule: 16-B blend
