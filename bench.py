@@ -5,8 +5,11 @@ Workload (BASELINE.json metric config — the 8-run merge the headline metric
 is quoted on): config 3 = 8 runs x ~1 GiB, 32 B keys / 1 KiB values, 50% key
 overlap, 5% tombstones, synthetic seeded runs (no network). A "step" = one
 full compaction of the 8-run set with inputs already resident in HBM
-(outputs land in HBM; PCIe-inclusive rates are reported in DESIGN.md, never
-as `value`).
+(outputs land in HBM). The default line also carries `end_to_end`: the
+PCIe-inclusive streamed-pinned-ingest rate measured beside the resident
+metric (never as `value` — SURVEY.md §8d), plus `cpu_baseline` (the C
+oracle restatement on the box's host cores). Other BASELINE shapes via
+--workload cfg2/cfg4/cfg5.
 
 Multi-GPU: the path shards as INDEPENDENT jobs — one
 compaction job per GPU (weak scaling), with the only collective an RCCL
