@@ -485,11 +485,12 @@ __global__ void k_corners(RunsDesc R, const uint64_t* pfx,
     }
 }
 
-template <int KB, bool TS>
-__global__ __launch_bounds__(CORANK_BLOCK) void k_corank(
+template <int KB, bool TS, int BLK = CORANK_BLOCK>
+__global__ __launch_bounds__(BLK) void k_corank(
     RunsDesc R, const uint64_t* pfx, const AuxT<KB, TS>* aux,
     const PairDesc* pairs, uint32_t n_pairs, uint64_t total_chunks,
     const uint64_t* corners, uint32_t* cr) {
+    constexpr int STEPS = CORANK_BLOCK_POS / BLK; /* positions/thread */
     __shared__ uint64_t s_pfx[CORANK_BLOCK_POS + 2];
     __shared__ uint32_t s_cr[CORANK_BLOCK_POS]; /* staged crossranks:
         [0..lenA) for run a, [lenA..lenA+lenB) for run b — each block's
@@ -528,9 +529,9 @@ __global__ __launch_bounds__(CORANK_BLOCK) void k_corank(
 
         /* stage both pfx segments (coalesced); sA = s_pfx[0..lenA),
          * sB = s_pfx[lenA..lenA+lenB) */
-        for (uint32_t u = threadIdx.x; u < lenA; u += CORANK_BLOCK)
+        for (uint32_t u = threadIdx.x; u < lenA; u += BLK)
             s_pfx[u] = gpa[iaS + u];
-        for (uint32_t u = threadIdx.x; u < lenB; u += CORANK_BLOCK)
+        for (uint32_t u = threadIdx.x; u < lenB; u += BLK)
             s_pfx[lenA + u] = gpb[ibS + u];
         __syncthreads();
         const uint64_t* sA = s_pfx;
@@ -538,7 +539,7 @@ __global__ __launch_bounds__(CORANK_BLOCK) void k_corank(
 
         /* per-thread sub-window: local diagonal search inside LDS */
         uint32_t L = lenA + lenB;
-        uint32_t d = threadIdx.x * CORANK_STEPS;
+        uint32_t d = threadIdx.x * STEPS;
         if (d < L) {
             uint32_t slo = d > lenB ? d - lenB : 0;
             uint32_t shi = d < lenA ? d : lenA;
@@ -554,7 +555,7 @@ __global__ __launch_bounds__(CORANK_BLOCK) void k_corank(
                     shi = mid;
             }
             uint32_t ja = slo, jb = d - slo;
-            uint32_t dend = d + CORANK_STEPS;
+            uint32_t dend = d + STEPS;
             if (dend > L) dend = L;
 
             for (uint32_t pos = d; pos < dend; pos++) {
@@ -613,9 +614,9 @@ __global__ __launch_bounds__(CORANK_BLOCK) void k_corank(
                             iaS;
             uint32_t* crb = cr + (uint64_t)sb * R.total + R.entry_base[b] +
                             ibS;
-            for (uint32_t u = threadIdx.x; u < lenA; u += CORANK_BLOCK)
+            for (uint32_t u = threadIdx.x; u < lenA; u += BLK)
                 cra[u] = s_cr[u];
-            for (uint32_t u = threadIdx.x; u < lenB; u += CORANK_BLOCK)
+            for (uint32_t u = threadIdx.x; u < lenB; u += BLK)
                 crb[u] = s_cr[lenA + u];
         }
         __syncthreads();
@@ -1979,6 +1980,9 @@ extern "C" int dbeel_gpu_job_run(dbeel_gpu_job* job, int keep_tombstones,
         uint32_t grid = pick_grid(n, 256);
         uint64_t cgrid = job->total_chunks;
         if (cgrid > 16384) cgrid = 16384;
+        int corank_wide = 0; /* A/B: 512-thread corank blocks */
+        if (const char* w = getenv("DBEEL_CORANK_BLOCK"))
+            corank_wide = atoi(w) >= 512;
         switch (job->aux_kind) {
 #define STAGE1(KB, TS)                                                      \
     case (KB == 12 ? 0 : (KB == 24 ? 1 : 2)):                               \
@@ -1994,12 +1998,23 @@ extern "C" int dbeel_gpu_job_run(dbeel_gpu_job* job, int keep_tombstones,
                                (const AuxT<KB, TS>*)job->d_aux,             \
                                (const PairDesc*)job->d_pairs, job->n_pairs, \
                                job->total_chunks, job->d_corners);          \
-            hipLaunchKernelGGL((k_corank<KB, TS>), dim3((uint32_t)cgrid),   \
-                               dim3(CORANK_BLOCK), 0, s, job->desc,         \
-                               job->d_pfx, (const AuxT<KB, TS>*)job->d_aux, \
-                               (const PairDesc*)job->d_pairs, job->n_pairs, \
-                               job->total_chunks, job->d_corners,           \
-                               job->d_cr);                                  \
+            if (corank_wide)                                                \
+                hipLaunchKernelGGL((k_corank<KB, TS, 512>),                 \
+                                   dim3((uint32_t)cgrid), dim3(512), 0, s,  \
+                                   job->desc, job->d_pfx,                   \
+                                   (const AuxT<KB, TS>*)job->d_aux,         \
+                                   (const PairDesc*)job->d_pairs,           \
+                                   job->n_pairs, job->total_chunks,         \
+                                   job->d_corners, job->d_cr);              \
+            else                                                            \
+                hipLaunchKernelGGL((k_corank<KB, TS, 256>),                 \
+                                   dim3((uint32_t)cgrid),                   \
+                                   dim3(CORANK_BLOCK), 0, s, job->desc,     \
+                                   job->d_pfx,                              \
+                                   (const AuxT<KB, TS>*)job->d_aux,         \
+                                   (const PairDesc*)job->d_pairs,           \
+                                   job->n_pairs, job->total_chunks,         \
+                                   job->d_corners, job->d_cr);              \
         }                                                                   \
         hipLaunchKernelGGL((k_rankreduce<KB, TS>), dim3(grid), dim3(256),   \
                            0, s, job->desc, job->d_pfx,                     \
