@@ -516,6 +516,24 @@ def test_sliced_compaction_parity(engine):
         assert gd == od
 
 
+def test_zero_data_run_with_entries_rejected(engine):
+    """A run whose index claims entries but whose data file is empty is
+    corrupt; both the normal pipeline and streamed ingest (which launches
+    the ranged prepare separately for zero-data runs) must flag it."""
+    from dbeel_amd.engine import DbeelGpuError
+
+    good = make_runs(2, 1_000, 16, 64, seed=3)
+    bad = [good[0], (np.zeros(0, dtype=np.uint8), good[1][1])]
+    with pytest.raises(DbeelGpuError) as ei:
+        engine.compact(bad, keep_tombstones=True, device=0)
+    assert ei.value.code == 2  # CORRUPT
+
+    with engine.Job(bad, device=0) as job:
+        with pytest.raises(DbeelGpuError) as ei:
+            job.ingest(bad)
+        assert ei.value.code == 2
+
+
 def test_ingest_rejects_corrupt_offsets(engine):
     """Streamed ingest with NON-MONOTONE index offsets (corrupt input)
     must report CORRUPT, never wrap a chunk copy or accept the data."""
